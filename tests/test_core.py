@@ -1,0 +1,228 @@
+"""Tests for the core domain: CreateAllocation sizing, zero-load path,
+transition penalties, system aggregation (mirrors reference pkg/core tests:
+allocation_test.go / system_test.go / server_test.go expectations)."""
+import math
+
+import pytest
+
+from inferno_amd.analyzer import (
+    Configuration,
+    DecodeParms as AD,
+    PrefillParms as AP,
+    QueueAnalyzer,
+    RequestSize,
+    ServiceParms,
+    TargetPerf,
+)
+from inferno_amd.config import (
+    ACCEL_PENALTY_FACTOR,
+    AllocationData,
+    ServerLoadSpec,
+    system_spec_from_json,
+    system_spec_to_json,
+)
+from inferno_amd.core import Allocation, System, create_allocation
+from tests.fixtures import make_spec
+
+
+def build_system(**kw):
+    spec = make_spec(**kw)
+    system, opt = System.from_spec(spec)
+    system.calculate()
+    return system, opt
+
+
+class TestCreateAllocation:
+    def test_feasible_allocation_sizing(self):
+        system, _ = build_system(n_servers=4, seed=1)
+        srv = system.servers["srv-0:ns"]
+        alloc = create_allocation(system, "srv-0:ns", "MI355X")
+        assert alloc is not None
+        assert alloc.accelerator == "MI355X"
+        assert alloc.num_replicas >= srv.min_num_replicas
+        # cost = acc.cost * numInstances * replicas
+        model = system.models[srv.model_name]
+        want_cost = (
+            system.accelerators["MI355X"].cost
+            * model.get_num_instances("MI355X")
+            * alloc.num_replicas
+        )
+        assert alloc.cost == pytest.approx(want_cost, rel=1e-6)
+        assert alloc.itl > 0 and alloc.ttft > 0
+        assert 0.0 <= alloc.rho <= 1.0
+
+    def test_replica_count_matches_manual_sizing(self):
+        system, _ = build_system(n_servers=2, seed=3)
+        srv = system.servers["srv-1:ns"]
+        perf = system.models[srv.model_name].get_perf_data("MI300X")
+        target = system.service_classes[srv.service_class_name].model_target(srv.model_name)
+        K = srv.load.avgOutTokens
+        N = max(perf.maxBatchSize * perf.atTokens // K, 1)
+        cfg = Configuration(
+            N,
+            10 * N,
+            ServiceParms(
+                AP(perf.prefillParms.gamma, perf.prefillParms.delta),
+                AD(perf.decodeParms.alpha, perf.decodeParms.beta),
+            ),
+        )
+        qa = QueueAnalyzer(cfg, RequestSize(srv.load.avgInTokens, K))
+        _, metrics, _ = qa.size(
+            TargetPerf(target_ttft=target.ttft, target_itl=target.itl, target_tps=target.tps)
+        )
+        want_replicas = max(
+            int(math.ceil((srv.load.arrivalRate / 60.0) / metrics.throughput)),
+            srv.min_num_replicas,
+        )
+        alloc = create_allocation(system, "srv-1:ns", "MI300X")
+        assert alloc is not None
+        assert alloc.num_replicas == want_replicas
+        assert alloc.batch_size == N
+
+    def test_unknown_lookups_return_none(self):
+        system, _ = build_system()
+        assert create_allocation(system, "nope", "MI355X") is None
+        assert create_allocation(system, "srv-0:ns", "H100") is None
+
+    def test_zero_load_with_min_replicas(self):
+        system, _ = build_system(n_servers=1, seed=5)
+        srv = system.servers["srv-0:ns"]
+        srv.load = ServerLoadSpec(arrivalRate=0.0, avgInTokens=0, avgOutTokens=0)
+        alloc = create_allocation(system, "srv-0:ns", "MI355X")
+        assert alloc is not None
+        assert alloc.num_replicas == 1
+        perf = system.models[srv.model_name].get_perf_data("MI355X")
+        assert alloc.batch_size == perf.maxBatchSize
+        assert alloc.itl == pytest.approx(perf.decodeParms.alpha + perf.decodeParms.beta)
+        assert alloc.ttft == pytest.approx(perf.prefillParms.gamma + perf.prefillParms.delta)
+        assert alloc.rho == 0.0
+
+    def test_zero_load_scale_to_zero(self):
+        system, _ = build_system(n_servers=1, seed=5, min_num_replicas=0)
+        srv = system.servers["srv-0:ns"]
+        srv.load = ServerLoadSpec(arrivalRate=0.0, avgInTokens=0, avgOutTokens=0)
+        alloc = create_allocation(system, "srv-0:ns", "MI355X")
+        assert alloc is not None
+        assert alloc.accelerator == ""
+        assert alloc.num_replicas == 0
+        assert alloc.cost == 0.0 and alloc.value == 0.0
+
+    def test_negative_load_rejected(self):
+        system, _ = build_system(n_servers=1)
+        system.servers["srv-0:ns"].load = ServerLoadSpec(arrivalRate=-1.0)
+        assert create_allocation(system, "srv-0:ns", "MI355X") is None
+
+    def test_infeasible_slo_returns_none(self):
+        system, _ = build_system(n_servers=1, seed=2)
+        srv = system.servers["srv-0:ns"]
+        # impossible ITL: below alpha on every accelerator
+        svc = system.service_classes[srv.service_class_name]
+        svc.targets[srv.model_name].itl = 1e-3
+        svc.targets[srv.model_name].ttft = 0.0
+        assert create_allocation(system, "srv-0:ns", "MI355X") is None
+
+
+class TestTransitionPenalty:
+    def test_same_accel_same_replicas(self):
+        a = Allocation(accelerator="g", num_replicas=2, cost=10.0)
+        b = Allocation(accelerator="g", num_replicas=2, cost=10.0)
+        assert a.transition_penalty(b) == 0.0
+
+    def test_same_accel_scale(self):
+        a = Allocation(accelerator="g", num_replicas=2, cost=10.0)
+        b = Allocation(accelerator="g", num_replicas=3, cost=15.0)
+        assert a.transition_penalty(b) == pytest.approx(5.0)
+
+    def test_accel_change(self):
+        a = Allocation(accelerator="g1", num_replicas=2, cost=10.0)
+        b = Allocation(accelerator="g2", num_replicas=1, cost=20.0)
+        want = ACCEL_PENALTY_FACTOR * 30.0 + 10.0
+        assert a.transition_penalty(b) == pytest.approx(want)
+
+    def test_candidate_value_is_penalty(self):
+        system, _ = build_system(n_servers=2, seed=7)
+        srv = system.servers["srv-0:ns"]
+        for name, alloc in srv.all_allocations.items():
+            assert alloc.value == pytest.approx(
+                srv.cur_allocation.transition_penalty(alloc), rel=1e-6
+            )
+
+
+class TestServer:
+    def test_keep_accelerator_restricts_candidates(self):
+        system, _ = build_system(n_servers=2, seed=4, keep_accelerator=True)
+        srv = system.servers["srv-0:ns"]
+        assert set(srv.all_allocations) <= {srv.cur_allocation.accelerator}
+
+    def test_all_candidates_without_keep(self):
+        system, _ = build_system(n_servers=2, seed=4)
+        srv = system.servers["srv-0:ns"]
+        assert set(srv.all_allocations) == {"MI355X", "MI325X", "MI300X"}
+
+    def test_saturated(self):
+        a = Allocation(accelerator="g", num_replicas=1, max_arrv_rate_per_replica=0.001)
+        # max RPM = 0.001*1000*60 = 60
+        assert not a.is_saturated(59.0)
+        assert a.is_saturated(61.0)
+
+
+class TestSystem:
+    def test_allocate_by_type(self):
+        system, opt = build_system(n_servers=4, seed=9)
+        from inferno_amd.solver import Manager, Optimizer
+
+        Manager(system, Optimizer(opt)).optimize()
+        agg = system.allocation_by_type
+        # manually accumulate
+        want: dict[str, tuple[int, float]] = {}
+        for srv in system.servers.values():
+            alloc = srv.allocation
+            if alloc is None:
+                continue
+            acc = system.accelerators[alloc.accelerator]
+            model = system.models[srv.model_name]
+            t = acc.type
+            c, cost = want.get(t, (0, 0.0))
+            want[t] = (
+                c + alloc.num_replicas * model.get_num_instances(acc.name) * acc.multiplicity,
+                cost + alloc.cost,
+            )
+        assert set(agg) == set(want)
+        for t, (c, cost) in want.items():
+            assert agg[t].count == c
+            assert agg[t].cost == pytest.approx(cost, rel=1e-6)
+
+    def test_generate_solution_includes_load(self):
+        system, opt = build_system(n_servers=2, seed=11)
+        from inferno_amd.solver import Manager, Optimizer
+
+        Manager(system, Optimizer(opt)).optimize()
+        sol = system.generate_solution()
+        assert set(sol) == set(system.servers)
+        for name, data in sol.items():
+            assert data.load is system.servers[name].load
+            assert data.numReplicas == system.servers[name].allocation.num_replicas
+
+    def test_spec_json_roundtrip(self):
+        spec = make_spec(n_servers=3, seed=13, capacity={"AMD-MI355X-288GB": 16})
+        doc = system_spec_to_json(spec)
+        back = system_spec_from_json(doc)
+        assert system_spec_to_json(back) == doc
+
+    def test_accelerator_power_model(self):
+        from inferno_amd.config import AcceleratorSpec, PowerSpec
+        from inferno_amd.core import Accelerator
+
+        acc = Accelerator(
+            AcceleratorSpec(
+                name="MI355X",
+                type="AMD-MI355X-288GB",
+                power=PowerSpec(idle=200, full=1400, midPower=1000, midUtil=0.5),
+            )
+        )
+        acc.calculate()
+        assert acc.power(0.0) == 200
+        assert acc.power(0.5) == 1000
+        assert acc.power(1.0) == 1400
+        assert acc.power(0.25) == pytest.approx(600)
+        assert acc.power(0.75) == pytest.approx(1200)
