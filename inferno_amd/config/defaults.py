@@ -16,6 +16,12 @@ MAX_QUEUE_TO_BATCH_RATIO = 10
 # Accelerator transition penalty factor (defaults.go:22)
 ACCEL_PENALTY_FACTOR = 0.1
 
+# Hard cap on queueing-chain batch states (max batch size N per cell).
+# The reference has no cap (its chain is an O(11*N) sequential recurrence);
+# this build caps N at the HIP kernel's LDS budget (WVA_MAX_N) and applies
+# the same cap on every backend so CPU and GPU sweeps agree exactly.
+MAX_BATCH_STATES = 8192
+
 # Service-class defaults (defaults.go:24-33)
 DEFAULT_SERVICE_CLASS_NAME = "Free"
 DEFAULT_LOW_PRIORITY = 100

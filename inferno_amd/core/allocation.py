@@ -21,7 +21,12 @@ from ..analyzer import (
     ServiceParms,
     TargetPerf,
 )
-from ..config import ACCEL_PENALTY_FACTOR, MAX_QUEUE_TO_BATCH_RATIO, AllocationData
+from ..config import (
+    ACCEL_PENALTY_FACTOR,
+    MAX_BATCH_STATES,
+    MAX_QUEUE_TO_BATCH_RATIO,
+    AllocationData,
+)
 
 
 @dataclass
@@ -191,6 +196,7 @@ def create_allocation(system, server_name: str, acc_name: str) -> Optional[Alloc
         N = server.max_batch_size
     else:
         N = max(perf.maxBatchSize * perf.atTokens // K, 1)
+    N = min(N, MAX_BATCH_STATES)
     max_queue = N * MAX_QUEUE_TO_BATCH_RATIO
 
     cfg = Configuration(

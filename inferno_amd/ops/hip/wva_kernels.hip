@@ -1,0 +1,552 @@
+// wva_kernels.hip — MI355X (gfx950/CDNA4) allocate-sweep + solver kernels.
+//
+// Replaces the scalar Go hot path of the reference autoscaler
+// (pkg/core/allocation.go:27-163 CreateAllocation, pkg/analyzer/
+// queueanalyzer.go:99-255 BuildModel/Analyze/Size, pkg/analyzer/
+// mm1modelstatedependent.go:70-116 computeProbabilities and
+// pkg/solver/solver.go:63-79 SolveUnlimited) with batched device kernels:
+//
+//   K1 wva_sweep : one 256-thread workgroup per (server, accelerator[, TP])
+//                  cell. Builds the state-dependent service-rate prefix in
+//                  LDS, then runs the TTFT and ITL lock-step bisections where
+//                  each chain evaluation is an O(N) strided pass + block
+//                  reductions; the 10N saturated queue states are an analytic
+//                  geometric tail (see inferno_amd/analyzer/queue.py for the
+//                  math). fp32 service rates (matching the reference's
+//                  float32 inputs), fp64 chain/log math (CDNA4 fp64 VALU).
+//   K2 wva_argmin: segmented argmin over the sweep output per server
+//                  (value = transition-penalty-adjusted cost), deterministic
+//                  lowest-cell-index tie-break.
+//
+// Built standalone with hipcc (no torch headers); exposed as a C ABI and
+// driven from Python via ctypes on torch tensors' device pointers.
+#include <hip/hip_runtime.h>
+#include <math.h>
+#include <stdint.h>
+
+#define WVA_BLOCK 256
+#define WVA_WAVE 64
+// max supported batch size (LDS: (N+1) doubles for the prefix + scratch)
+#define WVA_MAX_N 8192
+
+// bisection constants (ref pkg/analyzer/utils.go:8-9)
+#define WVA_TOL 1e-6
+#define WVA_MAX_ITERS 100
+// ref queueanalyzer.go:8-11
+#define WVA_EPSILON 1e-3
+#define WVA_STABILITY_FRACTION 0.1
+// ref pkg/config/defaults.go:22
+#define WVA_ACCEL_PENALTY 0.1f
+
+// ---------------------------------------------------------------------------
+// input/output SoA (all device pointers, one entry per cell unless noted)
+// ---------------------------------------------------------------------------
+struct WvaCellsIn {
+  // int32 inputs
+  const int *in_tok;          // avg input tokens
+  const int *out_tok;         // avg output tokens (K)
+  const int *batch_n;         // max batch size N for this cell (precomputed)
+  const int *min_replicas;    // server minNumReplicas
+  const int *perf_max_batch;  // perf.maxBatchSize (zero-load path batch)
+  const int *cur_replicas;    // current allocation replica count
+  const int *flags;           // bit0: cur accel == this accel; bit1: cur accel empty; bit2: has current alloc
+  // float32 inputs
+  const float *alpha;         // decode base (msec)
+  const float *beta;          // decode slope
+  const float *gamma;         // prefill base
+  const float *delta;         // prefill slope
+  const float *arrival_rate;  // req/min
+  const float *t_itl;         // SLO target ITL (msec), 0 = unconstrained
+  const float *t_ttft;        // SLO target TTFT (msec)
+  const float *t_tps;         // SLO target TPS (tok/sec)
+  const float *acc_cost;      // accelerator cost * numInstances (per replica)
+  const float *cur_cost;      // current allocation cost
+};
+
+struct WvaCellsOut {
+  uint8_t *feasible;   // 1 = allocation exists
+  uint8_t *zero_empty; // 1 = zero-load empty allocation (accelerator "")
+  int *num_replicas;
+  int *batch;
+  float *cost;
+  float *value;
+  float *itl;
+  float *ttft;
+  float *rho;
+  float *max_rate; // max arrival rate per replica (req/msec)
+};
+
+// ---------------------------------------------------------------------------
+// block-wide reductions over WVA_BLOCK threads (wave64-aware)
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ double block_reduce_max(double v, double *scratch) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int off = WVA_WAVE / 2; off > 0; off >>= 1)
+    v = fmax(v, __shfl_down(v, off, WVA_WAVE));
+  if ((tid & (WVA_WAVE - 1)) == 0) scratch[tid / WVA_WAVE] = v;
+  __syncthreads();
+  if (tid == 0) {
+    double m = scratch[0];
+#pragma unroll
+    for (int i = 1; i < WVA_BLOCK / WVA_WAVE; ++i) m = fmax(m, scratch[i]);
+    scratch[0] = m;
+  }
+  __syncthreads();
+  double r = scratch[0];
+  __syncthreads();
+  return r;
+}
+
+// reduces two values at once (sum of w and sum of n*w)
+__device__ __forceinline__ void block_reduce_sum2(double a, double b, double *scratch,
+                                                  double *outa, double *outb) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
+    a += __shfl_down(a, off, WVA_WAVE);
+    b += __shfl_down(b, off, WVA_WAVE);
+  }
+  if ((tid & (WVA_WAVE - 1)) == 0) {
+    scratch[2 * (tid / WVA_WAVE)] = a;
+    scratch[2 * (tid / WVA_WAVE) + 1] = b;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double sa = 0.0, sb = 0.0;
+#pragma unroll
+    for (int i = 0; i < WVA_BLOCK / WVA_WAVE; ++i) {
+      sa += scratch[2 * i];
+      sb += scratch[2 * i + 1];
+    }
+    scratch[0] = sa;
+    scratch[1] = sb;
+  }
+  __syncthreads();
+  *outa = scratch[0];
+  *outb = scratch[1];
+  __syncthreads();
+}
+
+// ---------------------------------------------------------------------------
+// queueing primitives (fp32 where the reference uses float32)
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float prefill_time_f(float gamma, float delta, int in_tok, float b) {
+  if (in_tok == 0) return 0.0f;
+  return gamma + delta * (float)in_tok * b;
+}
+__device__ __forceinline__ float decode_time_f(float alpha, float beta, float b) {
+  return alpha + beta * b;
+}
+
+// effective concurrency: invert avg service time to a batch level in [0, N]
+// (ref queueanalyzer.go:288-302)
+__device__ __forceinline__ double effective_concurrency(double serv_time, float gamma,
+                                                        float alpha, float delta, float beta,
+                                                        int in_tok, int out_tok, int N) {
+  double tokens = (double)(out_tok - 1);
+  double num = serv_time - ((double)gamma + (double)alpha * tokens);
+  double den = (double)delta * (double)in_tok + (double)beta * tokens;
+  double n;
+  if (den == 0.0)
+    n = (num > 0.0) ? (double)N : 0.0;
+  else
+    n = num / den;
+  return fmin(fmax(n, 0.0), (double)N);
+}
+
+struct ChainOut {
+  double throughput;  // req/msec
+  double wait;        // msec
+  double serv;        // msec
+  double in_servers;  // avg requests in service
+};
+
+// Solve the state-dependent chain at arrival rate lam. S = inclusive prefix
+// of log service rates in LDS (S[0]=0, S[n]=sum log s(1..n)); logsN = log s(N).
+// Every thread returns identical results.
+__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K,
+                               double *scratch) {
+  const int tid = threadIdx.x;
+  const double loglam = log(lam);
+  // pass 1: max over head terms t_n = n*loglam - S[n], n = 0..N
+  double tmax = -INFINITY;
+  for (int n = tid; n <= N; n += WVA_BLOCK) tmax = fmax(tmax, (double)n * loglam - S[n]);
+  const double m = block_reduce_max(tmax, scratch);
+  // pass 2: head sums of w = exp(t - m) and n*w
+  double hs = 0.0, hns = 0.0;
+  for (int n = tid; n <= N; n += WVA_BLOCK) {
+    double w = exp((double)n * loglam - S[n] - m);
+    hs += w;
+    hns += (double)n * w;
+  }
+  double head_sum, head_n_sum;
+  block_reduce_sum2(hs, hns, scratch, &head_sum, &head_n_sum);
+
+  // geometric tail n = N+1..K with ratio r = lam/s(N) (identical on all threads)
+  const double log_r = loglam - logsN;
+  const double r = exp(log_r);
+  const double wN = exp((double)N * loglam - S[N] - m);
+  const int Q = K - N;
+  double tail_sum = 0.0, tail_n_sum = 0.0, wK = (Q == 0) ? wN : 0.0;
+  if (Q > 0 && wN > 0.0) {
+    if (fabs(1.0 - r) < 1e-12) {
+      tail_sum = wN * (double)Q;
+      tail_n_sum = wN * ((double)Q * (double)N + (double)Q * (double)(Q + 1) * 0.5);
+      wK = wN;
+    } else {
+      const double rQ = exp((double)Q * log_r);
+      const double omr = 1.0 - r;
+      const double g = r * (1.0 - rQ) / omr;
+      const double jg = r * (1.0 - (double)(Q + 1) * rQ + (double)Q * rQ * r) / (omr * omr);
+      tail_sum = wN * g;
+      tail_n_sum = wN * ((double)N * g + jg);
+      wK = wN * rQ;
+    }
+  }
+  const double Z = head_sum + tail_sum;
+  const double pK = wK / Z;
+  const double avg_n_sys = (head_n_sum + tail_n_sum) / Z;
+  const double avg_n_serv = head_n_sum / Z + (1.0 - head_sum / Z) * (double)N;
+
+  ChainOut o;
+  o.throughput = lam * (1.0 - pK);
+  o.in_servers = avg_n_serv;
+  const double resp = avg_n_sys / o.throughput;
+  o.serv = avg_n_serv / o.throughput;
+  o.wait = fmax(resp - o.serv, 0.0);
+  return o;
+}
+
+// evaluate TTFT (kind 0) or ITL (kind 1) at lam — ref queueanalyzer.go:270-286
+__device__ double eval_metric(int kind, double lam, const double *S, double logsN, int N, int K,
+                              float gamma, float delta, float alpha, float beta, int in_tok,
+                              int out_tok, double *scratch) {
+  ChainOut c = chain_eval(lam, S, logsN, N, K, scratch);
+  double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+  if (kind == 0)
+    return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
+  return (double)decode_time_f(alpha, beta, (float)eff);
+}
+
+__device__ __forceinline__ bool within_tol(double x, double value) {
+  if (x == value) return true;
+  if (value == 0.0) return false;
+  return fabs((x - value) / value) <= WVA_TOL;
+}
+
+// binary search matching pkg/analyzer/utils.go:26-70 (whole block in lockstep;
+// every thread holds identical state). Returns x*, sets *ind to -1/0/+1.
+__device__ double bisect(int kind, double x_min, double x_max, double y_target, const double *S,
+                         double logsN, int N, int K, float gamma, float delta, float alpha,
+                         float beta, int in_tok, int out_tok, double *scratch, int *ind) {
+  double y_lo = eval_metric(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                            out_tok, scratch);
+  if (within_tol(y_lo, y_target)) { *ind = 0; return x_min; }
+  double y_hi = eval_metric(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                            out_tok, scratch);
+  if (within_tol(y_hi, y_target)) { *ind = 0; return x_max; }
+  const bool increasing = y_lo < y_hi;
+  if ((increasing && y_target < y_lo) || (!increasing && y_target > y_lo)) {
+    *ind = -1;
+    return x_min;
+  }
+  if ((increasing && y_target > y_hi) || (!increasing && y_target < y_hi)) {
+    *ind = +1;
+    return x_max;
+  }
+  double x_star = 0.5 * (x_min + x_max);
+  for (int it = 0; it < WVA_MAX_ITERS; ++it) {
+    x_star = 0.5 * (x_min + x_max);
+    double y_star = eval_metric(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                                out_tok, scratch);
+    if (within_tol(y_star, y_target)) break;
+    if ((increasing && y_target < y_star) || (!increasing && y_target > y_star))
+      x_max = x_star;
+    else
+      x_min = x_star;
+  }
+  *ind = 0;
+  return x_star;
+}
+
+// ---------------------------------------------------------------------------
+// K1: allocate-sweep — one workgroup per cell
+// dynamic LDS layout: S[0..maxN] (prefix), chunk[WVA_BLOCK+1] (scan carries),
+// scratch[2*WVA_BLOCK/WVA_WAVE ... 8] reductions reuse chunk tail.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in, WvaCellsOut out,
+                                                                  int n_cells) {
+  extern __shared__ double smem[];
+  const int cell = blockIdx.x;
+  if (cell >= n_cells) return;
+  const int tid = threadIdx.x;
+
+  const int in_tok = in.in_tok[cell];
+  const int out_tok = in.out_tok[cell];
+  const int N = in.batch_n[cell];
+  const int min_rep = in.min_replicas[cell];
+  const int flags = in.flags[cell];
+  const float alpha = in.alpha[cell];
+  const float beta = in.beta[cell];
+  const float gamma = in.gamma[cell];
+  const float delta = in.delta[cell];
+  const float arrival = in.arrival_rate[cell];
+  const float t_itl = in.t_itl[cell];
+  const float t_ttft = in.t_ttft[cell];
+  const float t_tps = in.t_tps[cell];
+  const float acc_cost = in.acc_cost[cell];
+  const float cur_cost = in.cur_cost[cell];
+  const int cur_rep = in.cur_replicas[cell];
+  const bool cur_same = flags & 1;
+  const bool cur_empty = flags & 2;
+  const bool has_cur = flags & 4;
+
+  // ---- zero-traffic path (ref allocation.go:73-75, 259-288) ----
+  if (arrival == 0.0f || out_tok == 0) {
+    if (tid == 0) {
+      if (min_rep == 0) {
+        // empty allocation: accelerator "", zero everything
+        out.feasible[cell] = 1;
+        out.zero_empty[cell] = 1;
+        out.num_replicas[cell] = 0;
+        out.batch[cell] = 0;
+        out.cost[cell] = 0.0f;
+        out.itl[cell] = 0.0f;
+        out.ttft[cell] = 0.0f;
+        out.rho[cell] = 0.0f;
+        out.max_rate[cell] = 0.0f;
+        float value = 0.0f;
+        if (has_cur) {
+          // penalty vs current: "same accelerator" iff current is also empty
+          if (cur_empty)
+            value = (cur_rep == 0) ? 0.0f : (0.0f - cur_cost);
+          else
+            value = WVA_ACCEL_PENALTY * cur_cost + (0.0f - cur_cost);
+        }
+        out.value[cell] = value;
+      } else {
+        int max_batch = in.perf_max_batch[cell];
+        // (host passes the server maxBatchSize override via perf_max_batch)
+        float cost = acc_cost * (float)min_rep;
+        float decode1 = alpha + beta;
+        float max_decode = alpha + beta * (float)max_batch;
+        float prefill1 = gamma + delta;
+        float max_serv = prefill1 + max_decode;
+        out.feasible[cell] = 1;
+        out.zero_empty[cell] = 0;
+        out.num_replicas[cell] = min_rep;
+        out.batch[cell] = max_batch;
+        out.cost[cell] = cost;
+        out.itl[cell] = decode1;
+        out.ttft[cell] = prefill1;
+        out.rho[cell] = 0.0f;
+        out.max_rate[cell] = (max_serv > 0.0f) ? (float)max_batch / max_serv : 0.0f;
+        float value = cost;
+        if (has_cur) {
+          if (cur_same)
+            value = (cur_rep == min_rep) ? 0.0f : (cost - cur_cost);
+          else
+            value = WVA_ACCEL_PENALTY * (cur_cost + cost) + (cost - cur_cost);
+        }
+        out.value[cell] = value;
+      }
+    }
+    return;
+  }
+
+  // ---- build state-dependent service rates + LDS log-prefix ----
+  // smem layout: S: [0..N] inclusive prefix of log s; then carries/scratch.
+  double *S = smem;                    // N+1 doubles
+  double *carry = smem + (N + 1);      // WVA_BLOCK + 1 doubles
+  double *scratch = carry + WVA_BLOCK + 1;  // 2*(WVA_BLOCK/WVA_WAVE) doubles
+
+  const int K = out_tok;  // avg output tokens (a.k.a. request length)
+  int num_decode = out_tok - 1;
+  if (in_tok == 0 && out_tok == 1) num_decode = 1;
+
+  // chunked inclusive scan over n = 1..N of log s(n):
+  // each thread scans a contiguous chunk, then block-scan of chunk totals.
+  const int chunk = (N + WVA_BLOCK - 1) / WVA_BLOCK;
+  const int n0 = tid * chunk + 1;
+  const int n1 = min(n0 + chunk - 1, N);
+  double local = 0.0;
+  for (int n = n0; n <= n1; ++n) {
+    float nf = (float)n;
+    float prefill = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok * nf);
+    float decode = alpha + beta * nf;
+    float s = nf / (prefill + (float)num_decode * decode);  // fp32 like the reference
+    local += log((double)s);
+    S[n] = local;  // chunk-local prefix for now
+  }
+  carry[tid + 1] = local;
+  if (tid == 0) carry[0] = 0.0;
+  __syncthreads();
+  // block scan of carries (Hillis-Steele over WVA_BLOCK+1 entries, thread 0..)
+  for (int off = 1; off <= WVA_BLOCK; off <<= 1) {
+    double v = (tid + 1 >= off) ? carry[tid + 1 - off] : 0.0;
+    __syncthreads();
+    carry[tid + 1] += v;
+    __syncthreads();
+  }
+  // add chunk offsets
+  const double offset = carry[tid];  // exclusive prefix of chunk totals
+  for (int n = n0; n <= n1; ++n) S[n] += offset;
+  if (tid == 0) S[0] = 0.0;
+  __syncthreads();
+
+  // s(1), s(N) in fp32 (rate-range bounds, ref queueanalyzer.go:117-119)
+  float prefill1 = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok);
+  float s1 = 1.0f / (prefill1 + (float)num_decode * (alpha + beta));
+  float prefillN = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok * (float)N);
+  float sN = (float)N / (prefillN + (float)num_decode * (alpha + beta * (float)N));
+  const double logsN = log((double)sN);
+
+  const double lam_min = (double)s1 * WVA_EPSILON;
+  const double lam_max = (double)sN * (1.0 - WVA_EPSILON);
+  const int Kstates = 11 * N;  // maxQueue (10N) + N  (ref allocation.go:87)
+
+  // ---- SLO sizing: TTFT and ITL bisections (ref queueanalyzer.go:185-255) --
+  int ind = 0;
+  bool feasible = true;
+  double lam_ttft = lam_max;
+  if (t_ttft > 0.0f) {
+    lam_ttft = bisect(0, lam_min, lam_max, (double)t_ttft, S, logsN, N, Kstates, gamma, delta,
+                      alpha, beta, in_tok, out_tok, scratch, &ind);
+    if (ind < 0) feasible = false;
+  }
+  double lam_itl = lam_max;
+  if (feasible && t_itl > 0.0f) {
+    lam_itl = bisect(1, lam_min, lam_max, (double)t_itl, S, logsN, N, Kstates, gamma, delta,
+                     alpha, beta, in_tok, out_tok, scratch, &ind);
+    if (ind < 0) feasible = false;
+  }
+  if (!feasible) {
+    if (tid == 0) {
+      out.feasible[cell] = 0;
+      out.zero_empty[cell] = 0;
+    }
+    return;
+  }
+  double lam_tps = (t_tps > 0.0f) ? lam_max * (1.0 - WVA_STABILITY_FRACTION) : lam_max;
+  double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
+
+  // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
+  ChainOut c = chain_eval(lam, S, logsN, N, Kstates, scratch);
+  const double rate_star = c.throughput * 1000.0;  // req/sec
+
+  double total_rate;  // req/sec (ref allocation.go:134-139)
+  if (t_tps == 0.0f)
+    total_rate = (double)arrival / 60.0;
+  else
+    total_rate = (double)t_tps / (double)K;
+  double reps_d = ceil(total_rate / rate_star);
+  if (!(reps_d > 0.0)) reps_d = 0.0;
+  if (reps_d > 2147483000.0) reps_d = 2147483000.0;
+  int num_replicas = (int)reps_d;
+  if (num_replicas < min_rep) num_replicas = min_rep;
+
+  const float cost = acc_cost * (float)num_replicas;
+
+  // ---- per-replica analyze (ref allocation.go:148-157) ----
+  const double rate = total_rate / (double)num_replicas;
+  ChainOut c2 = chain_eval(rate / 1000.0, S, logsN, N, Kstates, scratch);
+  const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+  const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
+  const float token_t = decode_time_f(alpha, beta, (float)eff);
+  double rho = c2.in_servers / (double)N;
+  rho = fmin(fmax(rho, 0.0), 1.0);
+
+  if (tid == 0) {
+    out.feasible[cell] = 1;
+    out.zero_empty[cell] = 0;
+    out.num_replicas[cell] = num_replicas;
+    out.batch[cell] = N;
+    out.cost[cell] = cost;
+    out.itl[cell] = token_t;
+    out.ttft[cell] = (float)c2.wait + prefill_t;
+    out.rho[cell] = (float)rho;
+    out.max_rate[cell] = (float)(rate_star / 1000.0);
+    // value = transition penalty from current allocation (ref server.go:55-67)
+    float value = cost;
+    if (has_cur) {
+      if (cur_same && !cur_empty)
+        value = (cur_rep == num_replicas) ? 0.0f : (cost - cur_cost);
+      else
+        value = WVA_ACCEL_PENALTY * (cur_cost + cost) + (cost - cur_cost);
+    }
+    out.value[cell] = value;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K2: segmented argmin per server over the sweep output.
+// cells must be grouped by server: seg_start[s]..seg_start[s+1] index cells.
+// Winner = min value among feasible cells; tie -> lowest cell index.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
+    const float *value, const uint8_t *feasible, const int *seg_start, int n_servers,
+    int *winner /* out: cell index or -1 per server */) {
+  const int srv = blockIdx.x;
+  if (srv >= n_servers) return;
+  const int lane = threadIdx.x;
+  const int beg = seg_start[srv], end = seg_start[srv + 1];
+  float best = INFINITY;
+  int best_i = -1;
+  for (int i = beg + lane; i < end; i += WVA_WAVE) {
+    if (!feasible[i]) continue;
+    float v = value[i];
+    if (best_i == -1 || v < best) {
+      best = v;
+      best_i = i;
+    }
+  }
+#pragma unroll
+  for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
+    float ov = __shfl_down(best, off, WVA_WAVE);
+    int oi = __shfl_down(best_i, off, WVA_WAVE);
+    if (oi != -1 && (best_i == -1 || ov < best || (ov == best && oi < best_i))) {
+      best = ov;
+      best_i = oi;
+    }
+  }
+  if (lane == 0) winner[srv] = best_i;
+}
+
+// ---------------------------------------------------------------------------
+// C ABI launchers
+// ---------------------------------------------------------------------------
+extern "C" int wva_sweep_launch(
+    int n_cells, int max_n, void *stream,
+    // int32 inputs
+    const int *in_tok, const int *out_tok, const int *batch_n, const int *min_replicas,
+    const int *perf_max_batch, const int *cur_replicas, const int *flags,
+    // f32 inputs
+    const float *alpha, const float *beta, const float *gamma, const float *delta,
+    const float *arrival_rate, const float *t_itl, const float *t_ttft, const float *t_tps,
+    const float *acc_cost, const float *cur_cost,
+    // outputs
+    uint8_t *feasible, uint8_t *zero_empty, int *num_replicas, int *batch, float *cost,
+    float *value, float *itl, float *ttft, float *rho, float *max_rate) {
+  if (n_cells <= 0) return 0;
+  if (max_n < 1 || max_n > WVA_MAX_N) return -2;
+  WvaCellsIn in = {in_tok, out_tok, batch_n, min_replicas, perf_max_batch, cur_replicas, flags,
+                   alpha, beta, gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
+                   cur_cost};
+  WvaCellsOut out = {feasible, zero_empty, num_replicas, batch, cost, value, itl, ttft, rho,
+                     max_rate};
+  size_t lds = (size_t)(max_n + 1 + WVA_BLOCK + 1 + 2 * (WVA_BLOCK / WVA_WAVE)) * sizeof(double);
+  hipLaunchKernelGGL(wva_sweep, dim3(n_cells), dim3(WVA_BLOCK), lds, (hipStream_t)stream, in,
+                     out, n_cells);
+  return (int)hipGetLastError();
+}
+
+extern "C" int wva_argmin_launch(int n_servers, void *stream, const float *value,
+                                 const uint8_t *feasible, const int *seg_start, int *winner) {
+  if (n_servers <= 0) return 0;
+  hipLaunchKernelGGL(wva_argmin, dim3(n_servers), dim3(WVA_WAVE), 0, (hipStream_t)stream, value,
+                     feasible, seg_start, n_servers, winner);
+  return (int)hipGetLastError();
+}
+
+extern "C" int wva_device_count(int *count) { return (int)hipGetDeviceCount(count); }

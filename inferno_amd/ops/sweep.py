@@ -1,0 +1,158 @@
+"""ctypes driver for the HIP allocate-sweep kernels.
+
+Loads the in-tree ``lib/libwva_hip.so`` and launches on torch's current HIP
+stream with raw device pointers. On a GPU box a missing/failed native library
+raises loudly (no silent eager fallback) — the CPU path must be requested
+explicitly via the engine's ``backend="cpu"``.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+from .build import LIB_PATH, build, needs_build
+
+MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
+
+_lib: Optional[ctypes.CDLL] = None
+
+
+class HipKernelError(RuntimeError):
+    pass
+
+
+def load_library(allow_build: bool = True) -> ctypes.CDLL:
+    """dlopen the kernel library (building it first if sources are newer)."""
+    global _lib
+    if _lib is not None:
+        return _lib
+    if needs_build():
+        if not allow_build:
+            raise HipKernelError(
+                f"HIP kernel library missing or stale at {LIB_PATH}; "
+                "run python -m inferno_amd.ops.build"
+            )
+        build()
+    lib = ctypes.CDLL(LIB_PATH)
+    lib.wva_sweep_launch.restype = ctypes.c_int
+    lib.wva_argmin_launch.restype = ctypes.c_int
+    lib.wva_device_count.restype = ctypes.c_int
+    _lib = lib
+    return lib
+
+
+def _ptr(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+@dataclass
+class SweepOutputs:
+    feasible: "object"  # torch uint8 [cells]
+    zero_empty: "object"
+    num_replicas: "object"  # int32
+    batch: "object"  # int32
+    cost: "object"  # float32
+    value: "object"
+    itl: "object"
+    ttft: "object"
+    rho: "object"
+    max_rate: "object"
+
+
+def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
+    """Launch the sweep kernel over cell SoA arrays (torch CPU tensors in,
+    results copied back to CPU tensors).
+
+    ``arrays`` keys (from engine.snapshot.build_cell_arrays): int32 tensors
+    in_tok,out_tok,batch_n,min_replicas,perf_max_batch,cur_replicas,flags and
+    float32 tensors alpha,beta,gamma,delta,arrival_rate,t_itl,t_ttft,t_tps,
+    acc_cost,cur_cost.
+    """
+    import torch
+
+    if not torch.cuda.is_available():
+        raise HipKernelError("run_sweep requires a GPU (use the CPU engine backend instead)")
+    lib = load_library()
+    n_cells = int(arrays["in_tok"].shape[0])
+    max_n = int(arrays["batch_n"].max().item()) if n_cells else 1
+    if max_n > MAX_N:
+        raise HipKernelError(f"batch size {max_n} exceeds kernel limit {MAX_N}")
+
+    dev = {k: v.to(device, non_blocking=True) for k, v in arrays.items()}
+    t = torch
+    out = SweepOutputs(
+        feasible=t.zeros(n_cells, dtype=t.uint8, device=device),
+        zero_empty=t.zeros(n_cells, dtype=t.uint8, device=device),
+        num_replicas=t.zeros(n_cells, dtype=t.int32, device=device),
+        batch=t.zeros(n_cells, dtype=t.int32, device=device),
+        cost=t.zeros(n_cells, dtype=t.float32, device=device),
+        value=t.zeros(n_cells, dtype=t.float32, device=device),
+        itl=t.zeros(n_cells, dtype=t.float32, device=device),
+        ttft=t.zeros(n_cells, dtype=t.float32, device=device),
+        rho=t.zeros(n_cells, dtype=t.float32, device=device),
+        max_rate=t.zeros(n_cells, dtype=t.float32, device=device),
+    )
+    stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+    rc = lib.wva_sweep_launch(
+        ctypes.c_int(n_cells),
+        ctypes.c_int(max(max_n, 1)),
+        stream,
+        _ptr(dev["in_tok"]),
+        _ptr(dev["out_tok"]),
+        _ptr(dev["batch_n"]),
+        _ptr(dev["min_replicas"]),
+        _ptr(dev["perf_max_batch"]),
+        _ptr(dev["cur_replicas"]),
+        _ptr(dev["flags"]),
+        _ptr(dev["alpha"]),
+        _ptr(dev["beta"]),
+        _ptr(dev["gamma"]),
+        _ptr(dev["delta"]),
+        _ptr(dev["arrival_rate"]),
+        _ptr(dev["t_itl"]),
+        _ptr(dev["t_ttft"]),
+        _ptr(dev["t_tps"]),
+        _ptr(dev["acc_cost"]),
+        _ptr(dev["cur_cost"]),
+        _ptr(out.feasible),
+        _ptr(out.zero_empty),
+        _ptr(out.num_replicas),
+        _ptr(out.batch),
+        _ptr(out.cost),
+        _ptr(out.value),
+        _ptr(out.itl),
+        _ptr(out.ttft),
+        _ptr(out.rho),
+        _ptr(out.max_rate),
+    )
+    if rc != 0:
+        raise HipKernelError(f"wva_sweep_launch failed with hipError {rc}")
+    return out
+
+
+def run_argmin(value, feasible, seg_start) -> "object":
+    """Segmented argmin over sweep outputs (device tensors). Returns winner
+    cell index (int32, -1 = no feasible candidate) per server, on device."""
+    import torch
+
+    lib = load_library()
+    n_servers = int(seg_start.shape[0]) - 1
+    winner = torch.full((n_servers,), -1, dtype=torch.int32, device=value.device)
+    stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+    rc = lib.wva_argmin_launch(
+        ctypes.c_int(n_servers),
+        stream,
+        _ptr(value),
+        _ptr(feasible),
+        _ptr(seg_start),
+        _ptr(winner),
+    )
+    if rc != 0:
+        raise HipKernelError(f"wva_argmin_launch failed with hipError {rc}")
+    return winner
+
+
+def library_loaded() -> bool:
+    return _lib is not None and os.path.exists(LIB_PATH)

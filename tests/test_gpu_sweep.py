@@ -1,0 +1,151 @@
+"""GPU differential tests: HIP wva_sweep/wva_argmin vs the CPU golden
+reference over randomized fleets. Require an MI355X (marked gpu)."""
+import math
+
+import numpy as np
+import pytest
+
+import torch
+
+from inferno_amd.config import OptimizerSpec, ServerLoadSpec
+from inferno_amd.core import System
+from inferno_amd.engine import SweepEngine
+from tests.fixtures import make_spec
+
+pytestmark = pytest.mark.gpu
+
+
+def build_pair(**kw):
+    a, opt = System.from_spec(make_spec(**kw))
+    b, _ = System.from_spec(make_spec(**kw))
+    return a, b, opt
+
+
+def assert_alloc_close(a, b, name, acc):
+    """a = golden CPU allocation, b = GPU allocation."""
+    assert a.accelerator == b.accelerator, f"{name}/{acc} accel"
+    if a.num_replicas != b.num_replicas:
+        # ceil boundary: allow off-by-one only when the sizing ratio is within
+        # bisection tolerance of an integer
+        assert abs(a.num_replicas - b.num_replicas) <= 1, f"{name}/{acc} replicas"
+        assert a.cost == pytest.approx(b.cost, rel=5e-2)
+        return
+    assert a.cost == pytest.approx(b.cost, rel=1e-5, abs=1e-4), f"{name}/{acc} cost"
+    assert a.value == pytest.approx(b.value, rel=1e-4, abs=1e-3), f"{name}/{acc} value"
+    assert a.itl == pytest.approx(b.itl, rel=1e-3, abs=1e-4), f"{name}/{acc} itl"
+    assert a.ttft == pytest.approx(b.ttft, rel=2e-3, abs=1e-3), f"{name}/{acc} ttft"
+    assert a.rho == pytest.approx(b.rho, rel=1e-3, abs=1e-5), f"{name}/{acc} rho"
+    assert a.max_arrv_rate_per_replica == pytest.approx(
+        b.max_arrv_rate_per_replica, rel=1e-4, abs=1e-9
+    )
+
+
+class TestSweepDifferential:
+    @pytest.mark.parametrize("seed", range(5))
+    def test_random_fleets(self, seed):
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=24, seed=100 + seed)
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        n_cells = n_match = 0
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map), f"feasibility mismatch for {name}"
+            for acc in a_map:
+                n_cells += 1
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+                if a_map[acc].num_replicas == b_map[acc].num_replicas:
+                    n_match += 1
+        # off-by-one replica flips must be rare
+        assert n_match >= n_cells - max(1, n_cells // 50)
+
+    def test_zero_load_cells(self):
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=4, seed=200)
+        for s in (cpu_sys, gpu_sys):
+            s.servers["srv-0:ns"].load = ServerLoadSpec(0.0, 0, 0)
+            s.servers["srv-1:ns"].load = ServerLoadSpec(0.0, 0, 0)
+            s.servers["srv-1:ns"].min_num_replicas = 0
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map)
+            for acc in a_map:
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+
+    def test_infeasible_slo(self):
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=2, seed=201)
+        for s in (cpu_sys, gpu_sys):
+            srv = s.servers["srv-0:ns"]
+            t = s.service_classes[srv.service_class_name].targets[srv.model_name]
+            t.itl = 1e-3
+            t.ttft = 0.0
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        assert cpu_sys.servers["srv-0:ns"].all_allocations == {}
+        assert gpu_sys.servers["srv-0:ns"].all_allocations == {}
+
+    def test_tps_target(self):
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=3, seed=202)
+        for s in (cpu_sys, gpu_sys):
+            for svc in s.service_classes.values():
+                for t in svc.targets.values():
+                    t.tps = 500.0
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map)
+            for acc in a_map:
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+
+    def test_large_batch_cells(self):
+        # push N near the LDS limit: small K with big maxBatchSize*atTokens
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=2, seed=203)
+        for s in (cpu_sys, gpu_sys):
+            for srv in s.servers.values():
+                srv.load = ServerLoadSpec(
+                    arrivalRate=120.0, avgInTokens=256, avgOutTokens=33
+                )
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map)
+            for acc in a_map:
+                assert b_map[acc].batch_size == a_map[acc].batch_size
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+
+
+class TestArgminKernel:
+    def test_winners_match_cpu_solve(self):
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=24, seed=300)
+        cpu_engine = SweepEngine(backend="cpu")
+        gpu_engine = SweepEngine(backend="gpu")
+        cpu_engine.solve(cpu_sys, opt)
+        gpu_engine.solve(gpu_sys, opt)
+        for name in cpu_sys.servers:
+            a = cpu_sys.servers[name].allocation
+            b = gpu_sys.servers[name].allocation
+            assert (a is None) == (b is None), name
+            if a is not None:
+                # values can tie or differ within fp tolerance; accept equal
+                # accelerator or equal value
+                if a.accelerator != b.accelerator:
+                    assert a.value == pytest.approx(b.value, rel=1e-4, abs=1e-3)
+
+    def test_native_library_is_loaded(self):
+        from inferno_amd.ops.sweep import library_loaded, load_library
+
+        load_library(allow_build=False)
+        assert library_loaded()
+
+
+class TestGraftEntry:
+    def test_smoke(self):
+        import __graft_entry__
+
+        __graft_entry__.smoke()
