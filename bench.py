@@ -43,6 +43,7 @@ def main():
     import torch.distributed as dist
 
     from inferno_amd.config import ServerLoadSpec
+    from inferno_amd.core import allocation_from_data
     from inferno_amd.core.system import System
     from inferno_amd.engine import SweepEngine
     from inferno_amd.parallel import ShardedSolver, shard_servers
@@ -94,7 +95,7 @@ def main():
         for name, data in result.solution.items():
             server = system.servers[name]
             server.spec.currentAlloc = data
-            server.apply_desired_alloc()
+            server.cur_allocation = allocation_from_data(data)
         return result
 
     cells_per_step = None

@@ -149,3 +149,39 @@ class TestGraftEntry:
         import __graft_entry__
 
         __graft_entry__.smoke()
+
+
+class TestFastPathGpu:
+    def test_fast_gpu_matches_fast_cpu(self):
+        from inferno_amd.engine import FastSweep
+
+        a, opt = System.from_spec(make_spec(n_servers=24, seed=400))
+        b, _ = System.from_spec(make_spec(n_servers=24, seed=400))
+        rec_c = FastSweep(a, backend="cpu").reconcile()
+        rec_g = FastSweep(b, backend="gpu").reconcile()
+        n = len(rec_c.acc_idx)
+        mismatch = 0
+        for i in range(n):
+            assert rec_c.acc_idx[i] == rec_g.acc_idx[i], i
+            if rec_c.num_replicas[i] != rec_g.num_replicas[i]:
+                assert abs(int(rec_c.num_replicas[i]) - int(rec_g.num_replicas[i])) <= 1
+                mismatch += 1
+                continue
+            assert rec_c.cost[i] == pytest.approx(rec_g.cost[i], rel=1e-5, abs=1e-3)
+            assert rec_c.itl[i] == pytest.approx(rec_g.itl[i], rel=1e-3, abs=1e-4)
+            assert rec_c.ttft[i] == pytest.approx(rec_g.ttft[i], rel=2e-3, abs=1e-3)
+        assert mismatch <= max(1, n // 20)
+
+    def test_sharded_fast_gpu_single_rank(self):
+        from inferno_amd.engine import SweepEngine
+        from inferno_amd.parallel import ShardedSolver
+
+        a, opt = System.from_spec(make_spec(n_servers=16, seed=401))
+        b, _ = System.from_spec(make_spec(n_servers=16, seed=401))
+        gpu = ShardedSolver(SweepEngine(backend="gpu"), fast=True).solve(a, opt)
+        cpu = ShardedSolver(SweepEngine(backend="cpu"), fast=True).solve(b, opt)
+        assert set(gpu.solution) == set(cpu.solution)
+        for name in gpu.solution:
+            g, c = gpu.solution[name], cpu.solution[name]
+            assert g.accelerator == c.accelerator
+            assert abs(g.numReplicas - c.numReplicas) <= 1
