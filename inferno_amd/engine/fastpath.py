@@ -197,6 +197,9 @@ class FastSweep:
         return self._reconcile_cpu()
 
     def _reconcile_gpu(self) -> WinnerRecord:
+        import os
+        import time
+
         import torch
 
         from ..ops.sweep import run_argmin, run_sweep
@@ -204,10 +207,23 @@ class FastSweep:
         n_srv = len(self.server_names)
         if self.n_cells == 0:
             return _empty_winner(n_srv)
+        trace = os.environ.get("INFERNO_TIMING", "") == "1"
+        if trace:
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
         arrays = self.cell_arrays()
+        if trace:
+            t1 = time.perf_counter()
         out = run_sweep(arrays, device=self.device)
         seg = torch.from_numpy(self.seg_start).to(self.device)
         winner = run_argmin(out.value, out.feasible, seg)
+        if trace:
+            torch.cuda.synchronize()
+            t2 = time.perf_counter()
+            print(
+                f"[timing] refresh={1e3*(t1-t0):.2f}ms sweep+argmin={1e3*(t2-t1):.2f}ms",
+                flush=True,
+            )
         # single D2H sync for everything needed to materialize winners
         w = winner.cpu().numpy()
         stack = torch.stack(
