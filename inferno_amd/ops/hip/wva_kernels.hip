@@ -6,14 +6,20 @@
 // mm1modelstatedependent.go:70-116 computeProbabilities and
 // pkg/solver/solver.go:63-79 SolveUnlimited) with batched device kernels:
 //
-//   K1 wva_sweep : one 256-thread workgroup per (server, accelerator[, TP])
-//                  cell. Builds the state-dependent service-rate prefix in
-//                  LDS, then runs the TTFT and ITL lock-step bisections where
-//                  each chain evaluation is an O(N) strided pass + block
-//                  reductions; the 10N saturated queue states are an analytic
-//                  geometric tail (see inferno_amd/analyzer/queue.py for the
-//                  math). fp32 service rates (matching the reference's
-//                  float32 inputs), fp64 chain/log math (CDNA4 fp64 VALU).
+//   K1 wva_sweep : ONE WAVE64 PER CELL (server, accelerator[, TP]). The
+//                  state-dependent service-rate log-prefix lives in LDS;
+//                  the TTFT and ITL SLO bisections run as ~10^2 lock-step
+//                  chain evaluations, each an O(N/64) strided pass whose
+//                  reductions are barrier-free __shfl_xor butterflies —
+//                  no __syncthreads in the hot loop (the wave is the CDNA
+//                  scheduling quantum; a 256-thread block design pays ~800
+//                  barriers per cell for the same math). The 10N saturated
+//                  queue states are an analytic geometric tail, making each
+//                  evaluation O(N) instead of the reference's O(11N)
+//                  sequential recurrence. fp32 service rates (matching the
+//                  reference's float32 inputs), fp64 chain/log math (CDNA4
+//                  fp64 VALU is 1/2 rate — cheap for this latency-bound
+//                  kernel).
 //   K2 wva_argmin: segmented argmin over the sweep output per server
 //                  (value = transition-penalty-adjusted cost), deterministic
 //                  lowest-cell-index tie-break.
@@ -24,9 +30,8 @@
 #include <math.h>
 #include <stdint.h>
 
-#define WVA_BLOCK 256
 #define WVA_WAVE 64
-// max supported batch size (LDS: (N+1) doubles for the prefix + scratch)
+// max supported batch size (LDS: (N+1) doubles for the log prefix)
 #define WVA_MAX_N 8192
 
 // bisection constants (ref pkg/analyzer/utils.go:8-9)
@@ -47,7 +52,7 @@ struct WvaCellsIn {
   const int *out_tok;         // avg output tokens (K)
   const int *batch_n;         // max batch size N for this cell (precomputed)
   const int *min_replicas;    // server minNumReplicas
-  const int *perf_max_batch;  // perf.maxBatchSize (zero-load path batch)
+  const int *perf_max_batch;  // perf.maxBatchSize or server override (zero-load path)
   const int *cur_replicas;    // current allocation replica count
   const int *flags;           // bit0: cur accel == this accel; bit1: cur accel empty; bit2: has current alloc
   // float32 inputs
@@ -77,55 +82,21 @@ struct WvaCellsOut {
 };
 
 // ---------------------------------------------------------------------------
-// block-wide reductions over WVA_BLOCK threads (wave64-aware)
+// wave-wide reductions: barrier-free __shfl_xor butterflies (all 64 lanes
+// end with the result)
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ double block_reduce_max(double v, double *scratch) {
-  const int tid = threadIdx.x;
+__device__ __forceinline__ double wave_max(double v) {
 #pragma unroll
-  for (int off = WVA_WAVE / 2; off > 0; off >>= 1)
-    v = fmax(v, __shfl_down(v, off, WVA_WAVE));
-  if ((tid & (WVA_WAVE - 1)) == 0) scratch[tid / WVA_WAVE] = v;
-  __syncthreads();
-  if (tid == 0) {
-    double m = scratch[0];
-#pragma unroll
-    for (int i = 1; i < WVA_BLOCK / WVA_WAVE; ++i) m = fmax(m, scratch[i]);
-    scratch[0] = m;
-  }
-  __syncthreads();
-  double r = scratch[0];
-  __syncthreads();
-  return r;
+  for (int off = WVA_WAVE / 2; off > 0; off >>= 1) v = fmax(v, __shfl_xor(v, off, WVA_WAVE));
+  return v;
 }
 
-// reduces two values at once (sum of w and sum of n*w)
-__device__ __forceinline__ void block_reduce_sum2(double a, double b, double *scratch,
-                                                  double *outa, double *outb) {
-  const int tid = threadIdx.x;
+__device__ __forceinline__ void wave_sum2(double &a, double &b) {
 #pragma unroll
   for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
-    a += __shfl_down(a, off, WVA_WAVE);
-    b += __shfl_down(b, off, WVA_WAVE);
+    a += __shfl_xor(a, off, WVA_WAVE);
+    b += __shfl_xor(b, off, WVA_WAVE);
   }
-  if ((tid & (WVA_WAVE - 1)) == 0) {
-    scratch[2 * (tid / WVA_WAVE)] = a;
-    scratch[2 * (tid / WVA_WAVE) + 1] = b;
-  }
-  __syncthreads();
-  if (tid == 0) {
-    double sa = 0.0, sb = 0.0;
-#pragma unroll
-    for (int i = 0; i < WVA_BLOCK / WVA_WAVE; ++i) {
-      sa += scratch[2 * i];
-      sb += scratch[2 * i + 1];
-    }
-    scratch[0] = sa;
-    scratch[1] = sb;
-  }
-  __syncthreads();
-  *outa = scratch[0];
-  *outb = scratch[1];
-  __syncthreads();
 }
 
 // ---------------------------------------------------------------------------
@@ -164,26 +135,24 @@ struct ChainOut {
 
 // Solve the state-dependent chain at arrival rate lam. S = inclusive prefix
 // of log service rates in LDS (S[0]=0, S[n]=sum log s(1..n)); logsN = log s(N).
-// Every thread returns identical results.
-__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K,
-                               double *scratch) {
-  const int tid = threadIdx.x;
+// Every lane returns identical results; no barriers (single wave).
+__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K) {
+  const int lane = threadIdx.x;
   const double loglam = log(lam);
   // pass 1: max over head terms t_n = n*loglam - S[n], n = 0..N
   double tmax = -INFINITY;
-  for (int n = tid; n <= N; n += WVA_BLOCK) tmax = fmax(tmax, (double)n * loglam - S[n]);
-  const double m = block_reduce_max(tmax, scratch);
+  for (int n = lane; n <= N; n += WVA_WAVE) tmax = fmax(tmax, (double)n * loglam - S[n]);
+  const double m = wave_max(tmax);
   // pass 2: head sums of w = exp(t - m) and n*w
-  double hs = 0.0, hns = 0.0;
-  for (int n = tid; n <= N; n += WVA_BLOCK) {
+  double head_sum = 0.0, head_n_sum = 0.0;
+  for (int n = lane; n <= N; n += WVA_WAVE) {
     double w = exp((double)n * loglam - S[n] - m);
-    hs += w;
-    hns += (double)n * w;
+    head_sum += w;
+    head_n_sum += (double)n * w;
   }
-  double head_sum, head_n_sum;
-  block_reduce_sum2(hs, hns, scratch, &head_sum, &head_n_sum);
+  wave_sum2(head_sum, head_n_sum);
 
-  // geometric tail n = N+1..K with ratio r = lam/s(N) (identical on all threads)
+  // geometric tail n = N+1..K with ratio r = lam/s(N) (identical on all lanes)
   const double log_r = loglam - logsN;
   const double r = exp(log_r);
   const double wN = exp((double)N * loglam - S[N] - m);
@@ -221,8 +190,8 @@ __device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N,
 // evaluate TTFT (kind 0) or ITL (kind 1) at lam — ref queueanalyzer.go:270-286
 __device__ double eval_metric(int kind, double lam, const double *S, double logsN, int N, int K,
                               float gamma, float delta, float alpha, float beta, int in_tok,
-                              int out_tok, double *scratch) {
-  ChainOut c = chain_eval(lam, S, logsN, N, K, scratch);
+                              int out_tok) {
+  ChainOut c = chain_eval(lam, S, logsN, N, K);
   double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   if (kind == 0)
     return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
@@ -235,16 +204,16 @@ __device__ __forceinline__ bool within_tol(double x, double value) {
   return fabs((x - value) / value) <= WVA_TOL;
 }
 
-// binary search matching pkg/analyzer/utils.go:26-70 (whole block in lockstep;
-// every thread holds identical state). Returns x*, sets *ind to -1/0/+1.
+// binary search matching pkg/analyzer/utils.go:26-70 (wave in lockstep;
+// every lane holds identical state). Returns x*, sets *ind to -1/0/+1.
 __device__ double bisect(int kind, double x_min, double x_max, double y_target, const double *S,
                          double logsN, int N, int K, float gamma, float delta, float alpha,
-                         float beta, int in_tok, int out_tok, double *scratch, int *ind) {
-  double y_lo = eval_metric(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                            out_tok, scratch);
+                         float beta, int in_tok, int out_tok, int *ind) {
+  double y_lo =
+      eval_metric(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
   if (within_tol(y_lo, y_target)) { *ind = 0; return x_min; }
-  double y_hi = eval_metric(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                            out_tok, scratch);
+  double y_hi =
+      eval_metric(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
   if (within_tol(y_hi, y_target)) { *ind = 0; return x_max; }
   const bool increasing = y_lo < y_hi;
   if ((increasing && y_target < y_lo) || (!increasing && y_target > y_lo)) {
@@ -258,8 +227,8 @@ __device__ double bisect(int kind, double x_min, double x_max, double y_target, 
   double x_star = 0.5 * (x_min + x_max);
   for (int it = 0; it < WVA_MAX_ITERS; ++it) {
     x_star = 0.5 * (x_min + x_max);
-    double y_star = eval_metric(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                                out_tok, scratch);
+    double y_star =
+        eval_metric(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
     if (within_tol(y_star, y_target)) break;
     if ((increasing && y_target < y_star) || (!increasing && y_target > y_star))
       x_max = x_star;
@@ -271,16 +240,15 @@ __device__ double bisect(int kind, double x_min, double x_max, double y_target, 
 }
 
 // ---------------------------------------------------------------------------
-// K1: allocate-sweep — one workgroup per cell
-// dynamic LDS layout: S[0..maxN] (prefix), chunk[WVA_BLOCK+1] (scan carries),
-// scratch[2*WVA_BLOCK/WVA_WAVE ... 8] reductions reuse chunk tail.
+// K1: allocate-sweep — one wave64 per cell.
+// dynamic LDS: S[0..maxN] inclusive prefix of log service rates.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in, WvaCellsOut out,
-                                                                  int n_cells) {
-  extern __shared__ double smem[];
+extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, WvaCellsOut out,
+                                                                 int n_cells) {
+  extern __shared__ double S[];
   const int cell = blockIdx.x;
   if (cell >= n_cells) return;
-  const int tid = threadIdx.x;
+  const int lane = threadIdx.x;
 
   const int in_tok = in.in_tok[cell];
   const int out_tok = in.out_tok[cell];
@@ -304,7 +272,7 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
 
   // ---- zero-traffic path (ref allocation.go:73-75, 259-288) ----
   if (arrival == 0.0f || out_tok == 0) {
-    if (tid == 0) {
+    if (lane == 0) {
       if (min_rep == 0) {
         // empty allocation: accelerator "", zero everything
         out.feasible[cell] = 1;
@@ -327,7 +295,6 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
         out.value[cell] = value;
       } else {
         int max_batch = in.perf_max_batch[cell];
-        // (host passes the server maxBatchSize override via perf_max_batch)
         float cost = acc_cost * (float)min_rep;
         float decode1 = alpha + beta;
         float max_decode = alpha + beta * (float)max_batch;
@@ -355,20 +322,15 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
     return;
   }
 
-  // ---- build state-dependent service rates + LDS log-prefix ----
-  // smem layout: S: [0..N] inclusive prefix of log s; then carries/scratch.
-  double *S = smem;                    // N+1 doubles
-  double *carry = smem + (N + 1);      // WVA_BLOCK + 1 doubles
-  double *scratch = carry + WVA_BLOCK + 1;  // 2*(WVA_BLOCK/WVA_WAVE) doubles
-
   const int K = out_tok;  // avg output tokens (a.k.a. request length)
   int num_decode = out_tok - 1;
   if (in_tok == 0 && out_tok == 1) num_decode = 1;
 
-  // chunked inclusive scan over n = 1..N of log s(n):
-  // each thread scans a contiguous chunk, then block-scan of chunk totals.
-  const int chunk = (N + WVA_BLOCK - 1) / WVA_BLOCK;
-  const int n0 = tid * chunk + 1;
+  // ---- build LDS log-prefix of service rates (chunked wave scan) ----
+  // each lane scans a contiguous chunk of n = 1..N; lane totals are scanned
+  // in registers with shfl_up; chunk offsets added back — no barriers.
+  const int chunk = (N + WVA_WAVE - 1) / WVA_WAVE;
+  const int n0 = lane * chunk + 1;
   const int n1 = min(n0 + chunk - 1, N);
   double local = 0.0;
   for (int n = n0; n <= n1; ++n) {
@@ -379,21 +341,21 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
     local += log((double)s);
     S[n] = local;  // chunk-local prefix for now
   }
-  carry[tid + 1] = local;
-  if (tid == 0) carry[0] = 0.0;
-  __syncthreads();
-  // block scan of carries (Hillis-Steele over WVA_BLOCK+1 entries, thread 0..)
-  for (int off = 1; off <= WVA_BLOCK; off <<= 1) {
-    double v = (tid + 1 >= off) ? carry[tid + 1 - off] : 0.0;
-    __syncthreads();
-    carry[tid + 1] += v;
-    __syncthreads();
+  // inclusive scan of lane totals -> exclusive offset for each lane
+  double incl = local;
+#pragma unroll
+  for (int off = 1; off < WVA_WAVE; off <<= 1) {
+    double v = __shfl_up(incl, off, WVA_WAVE);
+    if (lane >= off) incl += v;
   }
-  // add chunk offsets
-  const double offset = carry[tid];  // exclusive prefix of chunk totals
+  const double offset = incl - local;
   for (int n = n0; n <= n1; ++n) S[n] += offset;
-  if (tid == 0) S[0] = 0.0;
-  __syncthreads();
+  if (lane == 0) S[0] = 0.0;
+  // single-wave LDS visibility: lanes read other lanes' S entries next —
+  // the wave executes in lockstep and the compiler orders LDS ops with
+  // lgkmcnt waits; s_barrier is unnecessary with one wave, but keep the
+  // fence cheap and explicit:
+  __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0) & vmcnt(0)
 
   // s(1), s(N) in fp32 (rate-range bounds, ref queueanalyzer.go:117-119)
   float prefill1 = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok);
@@ -412,17 +374,17 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
   double lam_ttft = lam_max;
   if (t_ttft > 0.0f) {
     lam_ttft = bisect(0, lam_min, lam_max, (double)t_ttft, S, logsN, N, Kstates, gamma, delta,
-                      alpha, beta, in_tok, out_tok, scratch, &ind);
+                      alpha, beta, in_tok, out_tok, &ind);
     if (ind < 0) feasible = false;
   }
   double lam_itl = lam_max;
   if (feasible && t_itl > 0.0f) {
     lam_itl = bisect(1, lam_min, lam_max, (double)t_itl, S, logsN, N, Kstates, gamma, delta,
-                     alpha, beta, in_tok, out_tok, scratch, &ind);
+                     alpha, beta, in_tok, out_tok, &ind);
     if (ind < 0) feasible = false;
   }
   if (!feasible) {
-    if (tid == 0) {
+    if (lane == 0) {
       out.feasible[cell] = 0;
       out.zero_empty[cell] = 0;
     }
@@ -432,7 +394,7 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
   double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
 
   // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
-  ChainOut c = chain_eval(lam, S, logsN, N, Kstates, scratch);
+  ChainOut c = chain_eval(lam, S, logsN, N, Kstates);
   const double rate_star = c.throughput * 1000.0;  // req/sec
 
   double total_rate;  // req/sec (ref allocation.go:134-139)
@@ -450,14 +412,14 @@ extern "C" __global__ void __launch_bounds__(WVA_BLOCK) wva_sweep(WvaCellsIn in,
 
   // ---- per-replica analyze (ref allocation.go:148-157) ----
   const double rate = total_rate / (double)num_replicas;
-  ChainOut c2 = chain_eval(rate / 1000.0, S, logsN, N, Kstates, scratch);
+  ChainOut c2 = chain_eval(rate / 1000.0, S, logsN, N, Kstates);
   const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
   const float token_t = decode_time_f(alpha, beta, (float)eff);
   double rho = c2.in_servers / (double)N;
   rho = fmin(fmax(rho, 0.0), 1.0);
 
-  if (tid == 0) {
+  if (lane == 0) {
     out.feasible[cell] = 1;
     out.zero_empty[cell] = 0;
     out.num_replicas[cell] = num_replicas;
@@ -535,8 +497,8 @@ extern "C" int wva_sweep_launch(
                    cur_cost};
   WvaCellsOut out = {feasible, zero_empty, num_replicas, batch, cost, value, itl, ttft, rho,
                      max_rate};
-  size_t lds = (size_t)(max_n + 1 + WVA_BLOCK + 1 + 2 * (WVA_BLOCK / WVA_WAVE)) * sizeof(double);
-  hipLaunchKernelGGL(wva_sweep, dim3(n_cells), dim3(WVA_BLOCK), lds, (hipStream_t)stream, in,
+  size_t lds = (size_t)(max_n + 1) * sizeof(double);
+  hipLaunchKernelGGL(wva_sweep, dim3(n_cells), dim3(WVA_WAVE), lds, (hipStream_t)stream, in,
                      out, n_cells);
   return (int)hipGetLastError();
 }
