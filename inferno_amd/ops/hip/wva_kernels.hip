@@ -6,23 +6,24 @@
 // mm1modelstatedependent.go:70-116 computeProbabilities and
 // pkg/solver/solver.go:63-79 SolveUnlimited) with batched device kernels:
 //
-//   K1 wva_sweep : ONE WAVE64 PER CELL (server, accelerator[, TP]). The
-//                  state-dependent service-rate log-prefix lives in LDS;
-//                  the TTFT and ITL SLO bisections run as ~10^2 lock-step
-//                  chain evaluations, each an O(N/64) strided pass whose
-//                  reductions are barrier-free __shfl_xor butterflies —
-//                  no __syncthreads in the hot loop (the wave is the CDNA
-//                  scheduling quantum; a 256-thread block design pays ~800
-//                  barriers per cell for the same math). The 10N saturated
-//                  queue states are an analytic geometric tail, making each
-//                  evaluation O(N) instead of the reference's O(11N)
-//                  sequential recurrence. fp32 service rates (matching the
-//                  reference's float32 inputs), fp64 chain/log math (CDNA4
-//                  fp64 VALU is 1/2 rate — cheap for this latency-bound
-//                  kernel).
+//   K1 wva_sweep<NT>: one block per (server, accelerator[, TP]) cell,
+//       templated on block size and dispatched per N-bucket by the host:
+//         N <=  512  -> NT=64   (one wave; all reductions are barrier-free
+//                                __shfl_xor butterflies, scan carries in
+//                                registers — zero __syncthreads)
+//         N <= 2048  -> NT=256  (4 waves, hierarchical shfl+LDS reductions)
+//         N >  2048  -> NT=1024 (16 waves — the straggler cells' latency
+//                                sets the sweep's wall time, since ~all
+//                                cells are resident across 256 CUs at once)
+//       Per cell: fp32 state-dependent service rates (the reference's
+//       float32 inputs), fp64 log-prefix in LDS, then the TTFT and ITL SLO
+//       bisections as ~10^2 lock-step chain evaluations; each evaluation is
+//       an O(N/NT) strided pass + reductions, with the 10N saturated queue
+//       states folded into an analytic geometric tail (vs the reference's
+//       O(11N) sequential recurrence with overflow rescaling).
 //   K2 wva_argmin: segmented argmin over the sweep output per server
-//                  (value = transition-penalty-adjusted cost), deterministic
-//                  lowest-cell-index tie-break.
+//       (value = transition-penalty-adjusted cost), deterministic
+//       lowest-cell-index tie-break.
 //
 // Built standalone with hipcc (no torch headers); exposed as a C ABI and
 // driven from Python via ctypes on torch tensors' device pointers.
@@ -33,6 +34,9 @@
 #define WVA_WAVE 64
 // max supported batch size (LDS: (N+1) doubles for the log prefix)
 #define WVA_MAX_N 8192
+// N-bucket thresholds (host mirrors these in ops/sweep.py)
+#define WVA_N_SMALL 512
+#define WVA_N_MED 2048
 
 // bisection constants (ref pkg/analyzer/utils.go:8-9)
 #define WVA_TOL 1e-6
@@ -47,30 +51,28 @@
 // input/output SoA (all device pointers, one entry per cell unless noted)
 // ---------------------------------------------------------------------------
 struct WvaCellsIn {
-  // int32 inputs
-  const int *in_tok;          // avg input tokens
-  const int *out_tok;         // avg output tokens (K)
-  const int *batch_n;         // max batch size N for this cell (precomputed)
-  const int *min_replicas;    // server minNumReplicas
-  const int *perf_max_batch;  // perf.maxBatchSize or server override (zero-load path)
-  const int *cur_replicas;    // current allocation replica count
-  const int *flags;           // bit0: cur accel == this accel; bit1: cur accel empty; bit2: has current alloc
-  // float32 inputs
-  const float *alpha;         // decode base (msec)
-  const float *beta;          // decode slope
-  const float *gamma;         // prefill base
-  const float *delta;         // prefill slope
+  const int *in_tok;
+  const int *out_tok;
+  const int *batch_n;
+  const int *min_replicas;
+  const int *perf_max_batch;
+  const int *cur_replicas;
+  const int *flags;  // bit0: cur acc == this acc; bit1: cur acc empty; bit2: has cur
+  const float *alpha;
+  const float *beta;
+  const float *gamma;
+  const float *delta;
   const float *arrival_rate;  // req/min
-  const float *t_itl;         // SLO target ITL (msec), 0 = unconstrained
-  const float *t_ttft;        // SLO target TTFT (msec)
-  const float *t_tps;         // SLO target TPS (tok/sec)
-  const float *acc_cost;      // accelerator cost * numInstances (per replica)
-  const float *cur_cost;      // current allocation cost
+  const float *t_itl;
+  const float *t_ttft;
+  const float *t_tps;
+  const float *acc_cost;  // accelerator cost * numInstances (per replica)
+  const float *cur_cost;
 };
 
 struct WvaCellsOut {
-  uint8_t *feasible;   // 1 = allocation exists
-  uint8_t *zero_empty; // 1 = zero-load empty allocation (accelerator "")
+  uint8_t *feasible;
+  uint8_t *zero_empty;
   int *num_replicas;
   int *batch;
   float *cost;
@@ -78,24 +80,55 @@ struct WvaCellsOut {
   float *itl;
   float *ttft;
   float *rho;
-  float *max_rate; // max arrival rate per replica (req/msec)
+  float *max_rate;  // req/msec per replica
 };
 
 // ---------------------------------------------------------------------------
-// wave-wide reductions: barrier-free __shfl_xor butterflies (all 64 lanes
-// end with the result)
+// reductions templated on block size; scratch is LDS (>= 2*NT/WVA_WAVE doubles)
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ double wave_max(double v) {
+template <int NT>
+__device__ __forceinline__ double red_max(double v, double *scratch) {
 #pragma unroll
   for (int off = WVA_WAVE / 2; off > 0; off >>= 1) v = fmax(v, __shfl_xor(v, off, WVA_WAVE));
-  return v;
+  if constexpr (NT == WVA_WAVE) {
+    return v;
+  } else {
+    constexpr int NW = NT / WVA_WAVE;
+    const int tid = threadIdx.x;
+    __syncthreads();  // protect scratch from previous use
+    if ((tid & (WVA_WAVE - 1)) == 0) scratch[tid / WVA_WAVE] = v;
+    __syncthreads();
+    double m = scratch[0];
+#pragma unroll
+    for (int i = 1; i < NW; ++i) m = fmax(m, scratch[i]);
+    return m;
+  }
 }
 
-__device__ __forceinline__ void wave_sum2(double &a, double &b) {
+template <int NT>
+__device__ __forceinline__ void red_sum2(double &a, double &b, double *scratch) {
 #pragma unroll
   for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
     a += __shfl_xor(a, off, WVA_WAVE);
     b += __shfl_xor(b, off, WVA_WAVE);
+  }
+  if constexpr (NT != WVA_WAVE) {
+    constexpr int NW = NT / WVA_WAVE;
+    const int tid = threadIdx.x;
+    __syncthreads();
+    if ((tid & (WVA_WAVE - 1)) == 0) {
+      scratch[2 * (tid / WVA_WAVE)] = a;
+      scratch[2 * (tid / WVA_WAVE) + 1] = b;
+    }
+    __syncthreads();
+    double sa = 0.0, sb = 0.0;
+#pragma unroll
+    for (int i = 0; i < NW; ++i) {
+      sa += scratch[2 * i];
+      sb += scratch[2 * i + 1];
+    }
+    a = sa;
+    b = sb;
   }
 }
 
@@ -110,8 +143,7 @@ __device__ __forceinline__ float decode_time_f(float alpha, float beta, float b)
   return alpha + beta * b;
 }
 
-// effective concurrency: invert avg service time to a batch level in [0, N]
-// (ref queueanalyzer.go:288-302)
+// ref queueanalyzer.go:288-302
 __device__ __forceinline__ double effective_concurrency(double serv_time, float gamma,
                                                         float alpha, float delta, float beta,
                                                         int in_tok, int out_tok, int N) {
@@ -130,29 +162,29 @@ struct ChainOut {
   double throughput;  // req/msec
   double wait;        // msec
   double serv;        // msec
-  double in_servers;  // avg requests in service
+  double in_servers;
 };
 
 // Solve the state-dependent chain at arrival rate lam. S = inclusive prefix
-// of log service rates in LDS (S[0]=0, S[n]=sum log s(1..n)); logsN = log s(N).
-// Every lane returns identical results; no barriers (single wave).
-__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K) {
-  const int lane = threadIdx.x;
+// of log service rates in LDS; logsN = log s(N). All threads return
+// identical results.
+template <int NT>
+__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K,
+                               double *scratch) {
+  const int tid = threadIdx.x;
   const double loglam = log(lam);
-  // pass 1: max over head terms t_n = n*loglam - S[n], n = 0..N
   double tmax = -INFINITY;
-  for (int n = lane; n <= N; n += WVA_WAVE) tmax = fmax(tmax, (double)n * loglam - S[n]);
-  const double m = wave_max(tmax);
-  // pass 2: head sums of w = exp(t - m) and n*w
+  for (int n = tid; n <= N; n += NT) tmax = fmax(tmax, (double)n * loglam - S[n]);
+  const double m = red_max<NT>(tmax, scratch);
   double head_sum = 0.0, head_n_sum = 0.0;
-  for (int n = lane; n <= N; n += WVA_WAVE) {
+  for (int n = tid; n <= N; n += NT) {
     double w = exp((double)n * loglam - S[n] - m);
     head_sum += w;
     head_n_sum += (double)n * w;
   }
-  wave_sum2(head_sum, head_n_sum);
+  red_sum2<NT>(head_sum, head_n_sum, scratch);
 
-  // geometric tail n = N+1..K with ratio r = lam/s(N) (identical on all lanes)
+  // geometric tail n = N+1..K with ratio r = lam/s(N)
   const double log_r = loglam - logsN;
   const double r = exp(log_r);
   const double wN = exp((double)N * loglam - S[N] - m);
@@ -187,11 +219,11 @@ __device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N,
   return o;
 }
 
-// evaluate TTFT (kind 0) or ITL (kind 1) at lam — ref queueanalyzer.go:270-286
+template <int NT>
 __device__ double eval_metric(int kind, double lam, const double *S, double logsN, int N, int K,
                               float gamma, float delta, float alpha, float beta, int in_tok,
-                              int out_tok) {
-  ChainOut c = chain_eval(lam, S, logsN, N, K);
+                              int out_tok, double *scratch) {
+  ChainOut c = chain_eval<NT>(lam, S, logsN, N, K, scratch);
   double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   if (kind == 0)
     return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
@@ -204,16 +236,16 @@ __device__ __forceinline__ bool within_tol(double x, double value) {
   return fabs((x - value) / value) <= WVA_TOL;
 }
 
-// binary search matching pkg/analyzer/utils.go:26-70 (wave in lockstep;
-// every lane holds identical state). Returns x*, sets *ind to -1/0/+1.
+// binary search matching pkg/analyzer/utils.go:26-70 (block in lockstep)
+template <int NT>
 __device__ double bisect(int kind, double x_min, double x_max, double y_target, const double *S,
                          double logsN, int N, int K, float gamma, float delta, float alpha,
-                         float beta, int in_tok, int out_tok, int *ind) {
-  double y_lo =
-      eval_metric(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
+                         float beta, int in_tok, int out_tok, double *scratch, int *ind) {
+  double y_lo = eval_metric<NT>(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                                out_tok, scratch);
   if (within_tol(y_lo, y_target)) { *ind = 0; return x_min; }
-  double y_hi =
-      eval_metric(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
+  double y_hi = eval_metric<NT>(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                                out_tok, scratch);
   if (within_tol(y_hi, y_target)) { *ind = 0; return x_max; }
   const bool increasing = y_lo < y_hi;
   if ((increasing && y_target < y_lo) || (!increasing && y_target > y_lo)) {
@@ -227,8 +259,8 @@ __device__ double bisect(int kind, double x_min, double x_max, double y_target, 
   double x_star = 0.5 * (x_min + x_max);
   for (int it = 0; it < WVA_MAX_ITERS; ++it) {
     x_star = 0.5 * (x_min + x_max);
-    double y_star =
-        eval_metric(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta, in_tok, out_tok);
+    double y_star = eval_metric<NT>(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta,
+                                    in_tok, out_tok, scratch);
     if (within_tol(y_star, y_target)) break;
     if ((increasing && y_target < y_star) || (!increasing && y_target > y_star))
       x_max = x_star;
@@ -240,15 +272,19 @@ __device__ double bisect(int kind, double x_min, double x_max, double y_target, 
 }
 
 // ---------------------------------------------------------------------------
-// K1: allocate-sweep — one wave64 per cell.
-// dynamic LDS: S[0..maxN] inclusive prefix of log service rates.
+// K1: allocate-sweep — one block per cell; cell_ids selects this launch's
+// N-bucket (nullptr = identity).
+// dynamic LDS: S[0..maxN] prefix, then scratch[2*NT/64] for reductions.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, WvaCellsOut out,
-                                                                 int n_cells) {
-  extern __shared__ double S[];
-  const int cell = blockIdx.x;
-  if (cell >= n_cells) return;
-  const int lane = threadIdx.x;
+template <int NT>
+__global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out, int n_blocks,
+                                                  const int *cell_ids, int max_n) {
+  extern __shared__ double smem[];
+  if ((int)blockIdx.x >= n_blocks) return;
+  const int cell = cell_ids ? cell_ids[blockIdx.x] : (int)blockIdx.x;
+  const int tid = threadIdx.x;
+  double *S = smem;
+  double *scratch = smem + (max_n + 1);
 
   const int in_tok = in.in_tok[cell];
   const int out_tok = in.out_tok[cell];
@@ -272,9 +308,8 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
 
   // ---- zero-traffic path (ref allocation.go:73-75, 259-288) ----
   if (arrival == 0.0f || out_tok == 0) {
-    if (lane == 0) {
+    if (tid == 0) {
       if (min_rep == 0) {
-        // empty allocation: accelerator "", zero everything
         out.feasible[cell] = 1;
         out.zero_empty[cell] = 1;
         out.num_replicas[cell] = 0;
@@ -286,7 +321,6 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
         out.max_rate[cell] = 0.0f;
         float value = 0.0f;
         if (has_cur) {
-          // penalty vs current: "same accelerator" iff current is also empty
           if (cur_empty)
             value = (cur_rep == 0) ? 0.0f : (0.0f - cur_cost);
           else
@@ -322,15 +356,13 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
     return;
   }
 
-  const int K = out_tok;  // avg output tokens (a.k.a. request length)
+  const int K = out_tok;
   int num_decode = out_tok - 1;
   if (in_tok == 0 && out_tok == 1) num_decode = 1;
 
-  // ---- build LDS log-prefix of service rates (chunked wave scan) ----
-  // each lane scans a contiguous chunk of n = 1..N; lane totals are scanned
-  // in registers with shfl_up; chunk offsets added back — no barriers.
-  const int chunk = (N + WVA_WAVE - 1) / WVA_WAVE;
-  const int n0 = lane * chunk + 1;
+  // ---- build LDS log-prefix of service rates (chunked scan) ----
+  const int chunk = (N + NT - 1) / NT;
+  const int n0 = tid * chunk + 1;
   const int n1 = min(n0 + chunk - 1, N);
   double local = 0.0;
   for (int n = n0; n <= n1; ++n) {
@@ -341,21 +373,31 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
     local += log((double)s);
     S[n] = local;  // chunk-local prefix for now
   }
-  // inclusive scan of lane totals -> exclusive offset for each lane
+  // scan of per-thread chunk totals: intra-wave in registers...
   double incl = local;
 #pragma unroll
   for (int off = 1; off < WVA_WAVE; off <<= 1) {
     double v = __shfl_up(incl, off, WVA_WAVE);
-    if (lane >= off) incl += v;
+    if ((tid & (WVA_WAVE - 1)) >= off) incl += v;
   }
-  const double offset = incl - local;
+  double offset = incl - local;
+  if constexpr (NT != WVA_WAVE) {
+    // ...then cross-wave carries via scratch
+    constexpr int NW = NT / WVA_WAVE;
+    if ((tid & (WVA_WAVE - 1)) == WVA_WAVE - 1) scratch[tid / WVA_WAVE] = incl;
+    __syncthreads();
+    double carry = 0.0;
+    for (int w = 0; w < tid / WVA_WAVE; ++w) carry += scratch[w];
+    offset += carry;
+    static_assert(NW >= 1, "block must be at least one wave");
+  }
   for (int n = n0; n <= n1; ++n) S[n] += offset;
-  if (lane == 0) S[0] = 0.0;
-  // single-wave LDS visibility: lanes read other lanes' S entries next —
-  // the wave executes in lockstep and the compiler orders LDS ops with
-  // lgkmcnt waits; s_barrier is unnecessary with one wave, but keep the
-  // fence cheap and explicit:
-  __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0) & vmcnt(0)
+  if (tid == 0) S[0] = 0.0;
+  if constexpr (NT == WVA_WAVE) {
+    __builtin_amdgcn_s_waitcnt(0);  // single wave: order LDS writes before reads
+  } else {
+    __syncthreads();
+  }
 
   // s(1), s(N) in fp32 (rate-range bounds, ref queueanalyzer.go:117-119)
   float prefill1 = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok);
@@ -373,18 +415,18 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
   bool feasible = true;
   double lam_ttft = lam_max;
   if (t_ttft > 0.0f) {
-    lam_ttft = bisect(0, lam_min, lam_max, (double)t_ttft, S, logsN, N, Kstates, gamma, delta,
-                      alpha, beta, in_tok, out_tok, &ind);
+    lam_ttft = bisect<NT>(0, lam_min, lam_max, (double)t_ttft, S, logsN, N, Kstates, gamma,
+                          delta, alpha, beta, in_tok, out_tok, scratch, &ind);
     if (ind < 0) feasible = false;
   }
   double lam_itl = lam_max;
   if (feasible && t_itl > 0.0f) {
-    lam_itl = bisect(1, lam_min, lam_max, (double)t_itl, S, logsN, N, Kstates, gamma, delta,
-                     alpha, beta, in_tok, out_tok, &ind);
+    lam_itl = bisect<NT>(1, lam_min, lam_max, (double)t_itl, S, logsN, N, Kstates, gamma, delta,
+                         alpha, beta, in_tok, out_tok, scratch, &ind);
     if (ind < 0) feasible = false;
   }
   if (!feasible) {
-    if (lane == 0) {
+    if (tid == 0) {
       out.feasible[cell] = 0;
       out.zero_empty[cell] = 0;
     }
@@ -394,7 +436,7 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
   double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
 
   // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
-  ChainOut c = chain_eval(lam, S, logsN, N, Kstates);
+  ChainOut c = chain_eval<NT>(lam, S, logsN, N, Kstates, scratch);
   const double rate_star = c.throughput * 1000.0;  // req/sec
 
   double total_rate;  // req/sec (ref allocation.go:134-139)
@@ -412,14 +454,14 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
 
   // ---- per-replica analyze (ref allocation.go:148-157) ----
   const double rate = total_rate / (double)num_replicas;
-  ChainOut c2 = chain_eval(rate / 1000.0, S, logsN, N, Kstates);
+  ChainOut c2 = chain_eval<NT>(rate / 1000.0, S, logsN, N, Kstates, scratch);
   const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
   const float token_t = decode_time_f(alpha, beta, (float)eff);
   double rho = c2.in_servers / (double)N;
   rho = fmin(fmax(rho, 0.0), 1.0);
 
-  if (lane == 0) {
+  if (tid == 0) {
     out.feasible[cell] = 1;
     out.zero_empty[cell] = 0;
     out.num_replicas[cell] = num_replicas;
@@ -429,7 +471,6 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
     out.ttft[cell] = (float)c2.wait + prefill_t;
     out.rho[cell] = (float)rho;
     out.max_rate[cell] = (float)(rate_star / 1000.0);
-    // value = transition penalty from current allocation (ref server.go:55-67)
     float value = cost;
     if (has_cur) {
       if (cur_same && !cur_empty)
@@ -443,12 +484,10 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_sweep(WvaCellsIn in, 
 
 // ---------------------------------------------------------------------------
 // K2: segmented argmin per server over the sweep output.
-// cells must be grouped by server: seg_start[s]..seg_start[s+1] index cells.
-// Winner = min value among feasible cells; tie -> lowest cell index.
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
     const float *value, const uint8_t *feasible, const int *seg_start, int n_servers,
-    int *winner /* out: cell index or -1 per server */) {
+    int *winner) {
   const int srv = blockIdx.x;
   if (srv >= n_servers) return;
   const int lane = threadIdx.x;
@@ -478,29 +517,58 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
 // ---------------------------------------------------------------------------
 // C ABI launchers
 // ---------------------------------------------------------------------------
-extern "C" int wva_sweep_launch(
-    int n_cells, int max_n, void *stream,
-    // int32 inputs
+extern "C" int wva_sweep_launch_bucket(
+    int n_blocks, int max_n, int nt, const int *cell_ids, void *stream,
     const int *in_tok, const int *out_tok, const int *batch_n, const int *min_replicas,
     const int *perf_max_batch, const int *cur_replicas, const int *flags,
-    // f32 inputs
     const float *alpha, const float *beta, const float *gamma, const float *delta,
     const float *arrival_rate, const float *t_itl, const float *t_ttft, const float *t_tps,
     const float *acc_cost, const float *cur_cost,
-    // outputs
     uint8_t *feasible, uint8_t *zero_empty, int *num_replicas, int *batch, float *cost,
     float *value, float *itl, float *ttft, float *rho, float *max_rate) {
-  if (n_cells <= 0) return 0;
+  if (n_blocks <= 0) return 0;
   if (max_n < 1 || max_n > WVA_MAX_N) return -2;
   WvaCellsIn in = {in_tok, out_tok, batch_n, min_replicas, perf_max_batch, cur_replicas, flags,
                    alpha, beta, gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
                    cur_cost};
   WvaCellsOut out = {feasible, zero_empty, num_replicas, batch, cost, value, itl, ttft, rho,
                      max_rate};
-  size_t lds = (size_t)(max_n + 1) * sizeof(double);
-  hipLaunchKernelGGL(wva_sweep, dim3(n_cells), dim3(WVA_WAVE), lds, (hipStream_t)stream, in,
-                     out, n_cells);
+  size_t lds = (size_t)(max_n + 1 + 2 * (1024 / WVA_WAVE)) * sizeof(double);
+  switch (nt) {
+    case 64:
+      hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,
+                         in, out, n_blocks, cell_ids, max_n);
+      break;
+    case 256:
+      hipLaunchKernelGGL(wva_sweep_t<256>, dim3(n_blocks), dim3(256), lds, (hipStream_t)stream,
+                         in, out, n_blocks, cell_ids, max_n);
+      break;
+    case 1024:
+      hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(n_blocks), dim3(1024), lds,
+                         (hipStream_t)stream, in, out, n_blocks, cell_ids, max_n);
+      break;
+    default:
+      return -3;
+  }
   return (int)hipGetLastError();
+}
+
+// single-bucket convenience wrapper (whole sweep with one block size)
+extern "C" int wva_sweep_launch(
+    int n_cells, int max_n, void *stream,
+    const int *in_tok, const int *out_tok, const int *batch_n, const int *min_replicas,
+    const int *perf_max_batch, const int *cur_replicas, const int *flags,
+    const float *alpha, const float *beta, const float *gamma, const float *delta,
+    const float *arrival_rate, const float *t_itl, const float *t_ttft, const float *t_tps,
+    const float *acc_cost, const float *cur_cost,
+    uint8_t *feasible, uint8_t *zero_empty, int *num_replicas, int *batch, float *cost,
+    float *value, float *itl, float *ttft, float *rho, float *max_rate) {
+  int nt = (max_n <= WVA_N_SMALL) ? 64 : (max_n <= WVA_N_MED ? 256 : 1024);
+  return wva_sweep_launch_bucket(n_cells, max_n, nt, nullptr, stream, in_tok, out_tok, batch_n,
+                                 min_replicas, perf_max_batch, cur_replicas, flags, alpha, beta,
+                                 gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
+                                 cur_cost, feasible, zero_empty, num_replicas, batch, cost,
+                                 value, itl, ttft, rho, max_rate);
 }
 
 extern "C" int wva_argmin_launch(int n_servers, void *stream, const float *value,

@@ -15,6 +15,12 @@ from typing import Optional
 from .build import LIB_PATH, build, needs_build
 
 MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
+# N-bucket thresholds (must match WVA_N_SMALL / WVA_N_MED): cells are
+# dispatched to 64/256/1024-thread blocks by batch size — the straggler
+# (large-N) cells' latency sets the sweep wall time, so they get wide blocks,
+# while small cells run one barrier-free wave each.
+N_SMALL = 512
+N_MED = 2048
 
 _lib: Optional[ctypes.CDLL] = None
 
@@ -37,6 +43,7 @@ def load_library(allow_build: bool = True) -> ctypes.CDLL:
         build()
     lib = ctypes.CDLL(LIB_PATH)
     lib.wva_sweep_launch.restype = ctypes.c_int
+    lib.wva_sweep_launch_bucket.restype = ctypes.c_int
     lib.wva_argmin_launch.restype = ctypes.c_int
     lib.wva_device_count.restype = ctypes.c_int
     _lib = lib
@@ -95,40 +102,63 @@ def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
         max_rate=t.zeros(n_cells, dtype=t.float32, device=device),
     )
     stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
-    rc = lib.wva_sweep_launch(
-        ctypes.c_int(n_cells),
-        ctypes.c_int(max(max_n, 1)),
-        stream,
-        _ptr(dev["in_tok"]),
-        _ptr(dev["out_tok"]),
-        _ptr(dev["batch_n"]),
-        _ptr(dev["min_replicas"]),
-        _ptr(dev["perf_max_batch"]),
-        _ptr(dev["cur_replicas"]),
-        _ptr(dev["flags"]),
-        _ptr(dev["alpha"]),
-        _ptr(dev["beta"]),
-        _ptr(dev["gamma"]),
-        _ptr(dev["delta"]),
-        _ptr(dev["arrival_rate"]),
-        _ptr(dev["t_itl"]),
-        _ptr(dev["t_ttft"]),
-        _ptr(dev["t_tps"]),
-        _ptr(dev["acc_cost"]),
-        _ptr(dev["cur_cost"]),
-        _ptr(out.feasible),
-        _ptr(out.zero_empty),
-        _ptr(out.num_replicas),
-        _ptr(out.batch),
-        _ptr(out.cost),
-        _ptr(out.value),
-        _ptr(out.itl),
-        _ptr(out.ttft),
-        _ptr(out.rho),
-        _ptr(out.max_rate),
-    )
-    if rc != 0:
-        raise HipKernelError(f"wva_sweep_launch failed with hipError {rc}")
+
+    # partition cells into N-buckets so stragglers get wide blocks
+    batch_cpu = arrays["batch_n"]
+    buckets = []  # (nt, ids tensor or None, bucket_max_n)
+    small = batch_cpu <= N_SMALL
+    med = (batch_cpu > N_SMALL) & (batch_cpu <= N_MED)
+    large = batch_cpu > N_MED
+    if bool(small.all()):
+        buckets.append((64, None, int(max_n)))
+    elif bool(med.all()):
+        buckets.append((256, None, int(max_n)))
+    elif bool(large.all()):
+        buckets.append((1024, None, int(max_n)))
+    else:
+        for nt, mask in ((64, small), (256, med), (1024, large)):
+            ids = mask.nonzero(as_tuple=False).flatten().to(t.int32)
+            if ids.numel():
+                buckets.append((nt, ids.to(device), int(batch_cpu[mask].max().item())))
+
+    for nt, ids, bmax in buckets:
+        n_blocks = n_cells if ids is None else int(ids.numel())
+        rc = lib.wva_sweep_launch_bucket(
+            ctypes.c_int(n_blocks),
+            ctypes.c_int(max(bmax, 1)),
+            ctypes.c_int(nt),
+            _ptr(ids) if ids is not None else None,
+            stream,
+            _ptr(dev["in_tok"]),
+            _ptr(dev["out_tok"]),
+            _ptr(dev["batch_n"]),
+            _ptr(dev["min_replicas"]),
+            _ptr(dev["perf_max_batch"]),
+            _ptr(dev["cur_replicas"]),
+            _ptr(dev["flags"]),
+            _ptr(dev["alpha"]),
+            _ptr(dev["beta"]),
+            _ptr(dev["gamma"]),
+            _ptr(dev["delta"]),
+            _ptr(dev["arrival_rate"]),
+            _ptr(dev["t_itl"]),
+            _ptr(dev["t_ttft"]),
+            _ptr(dev["t_tps"]),
+            _ptr(dev["acc_cost"]),
+            _ptr(dev["cur_cost"]),
+            _ptr(out.feasible),
+            _ptr(out.zero_empty),
+            _ptr(out.num_replicas),
+            _ptr(out.batch),
+            _ptr(out.cost),
+            _ptr(out.value),
+            _ptr(out.itl),
+            _ptr(out.ttft),
+            _ptr(out.rho),
+            _ptr(out.max_rate),
+        )
+        if rc != 0:
+            raise HipKernelError(f"wva_sweep_launch_bucket(nt={nt}) failed with hipError {rc}")
     return out
 
 
