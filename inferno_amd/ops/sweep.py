@@ -121,14 +121,23 @@ def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
             if ids.numel():
                 buckets.append((nt, ids.to(device), int(batch_cpu[mask].max().item())))
 
-    for nt, ids, bmax in buckets:
+    # overlap the bucket launches on separate HIP streams: wall time becomes
+    # the straggler bucket's latency instead of the sum of all three
+    main_stream = torch.cuda.current_stream()
+    side = [torch.cuda.Stream() for _ in range(max(len(buckets) - 1, 0))]
+    for i, (nt, ids, bmax) in enumerate(buckets):
+        if i == 0:
+            cur = main_stream
+        else:
+            cur = side[i - 1]
+            cur.wait_stream(main_stream)  # order after the H2D uploads
         n_blocks = n_cells if ids is None else int(ids.numel())
         rc = lib.wva_sweep_launch_bucket(
             ctypes.c_int(n_blocks),
             ctypes.c_int(max(bmax, 1)),
             ctypes.c_int(nt),
             _ptr(ids) if ids is not None else None,
-            stream,
+            ctypes.c_void_p(cur.cuda_stream),
             _ptr(dev["in_tok"]),
             _ptr(dev["out_tok"]),
             _ptr(dev["batch_n"]),
@@ -159,6 +168,8 @@ def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
         )
         if rc != 0:
             raise HipKernelError(f"wva_sweep_launch_bucket(nt={nt}) failed with hipError {rc}")
+    for s in side:
+        main_stream.wait_stream(s)
     return out
 
 
