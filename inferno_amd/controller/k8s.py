@@ -1,0 +1,204 @@
+"""Minimal Kubernetes client layer.
+
+``HttpKube`` talks to the API server over REST (httpx, in-cluster service
+account or kubeconfig-provided token/CA) for the resources the controller
+needs: VariantAutoscalings (CRD), Deployments, ConfigMaps, Leases (leader
+election). ``InMemoryKube`` is the envtest-style fake used by the test
+suites (SURVEY.md section 4 tier 2/3).
+"""
+from __future__ import annotations
+
+import copy
+import os
+from dataclasses import dataclass, field
+from typing import Any, Optional, Protocol
+
+from ..api import v1alpha1 as api
+
+
+@dataclass
+class Deployment:
+    name: str
+    namespace: str
+    replicas: int = 1  # spec.replicas
+    status_replicas: int = 0  # status.replicas
+    uid: str = ""
+    labels: dict[str, str] = field(default_factory=dict)
+
+
+class KubeClient(Protocol):  # pragma: no cover - protocol
+    def list_variantautoscalings(self) -> list[api.VariantAutoscaling]: ...
+
+    def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]: ...
+
+    def get_deployment(self, namespace: str, name: str) -> Optional[Deployment]: ...
+
+    def update_va_status(self, va: api.VariantAutoscaling) -> None: ...
+
+    def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None: ...
+
+
+class InMemoryKube:
+    """In-memory fake of the API server (test double)."""
+
+    def __init__(self) -> None:
+        self.vas: dict[tuple[str, str], api.VariantAutoscaling] = {}
+        self.configmaps: dict[tuple[str, str], dict[str, str]] = {}
+        self.deployments: dict[tuple[str, str], Deployment] = {}
+        self.status_updates: list[dict[str, Any]] = []
+
+    # -- setup helpers -------------------------------------------------
+    def add_va(self, va: api.VariantAutoscaling) -> None:
+        self.vas[(va.namespace, va.name)] = va
+
+    def add_configmap(self, namespace: str, name: str, data: dict[str, str]) -> None:
+        self.configmaps[(namespace, name)] = dict(data)
+
+    def add_deployment(self, deploy: Deployment) -> None:
+        self.deployments[(deploy.namespace, deploy.name)] = deploy
+
+    # -- KubeClient ----------------------------------------------------
+    def list_variantautoscalings(self) -> list[api.VariantAutoscaling]:
+        return [copy.deepcopy(v) for v in self.vas.values()]
+
+    def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]:
+        cm = self.configmaps.get((namespace, name))
+        return dict(cm) if cm is not None else None
+
+    def get_deployment(self, namespace: str, name: str) -> Optional[Deployment]:
+        return self.deployments.get((namespace, name))
+
+    def update_va_status(self, va: api.VariantAutoscaling) -> None:
+        key = (va.namespace, va.name)
+        if key not in self.vas:
+            raise KeyError(f"VariantAutoscaling {key} not found")
+        stored = self.vas[key]
+        stored.status = copy.deepcopy(va.status)
+        self.status_updates.append(api.va_to_json(va)["status"])
+
+    def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
+        ref = {
+            "apiVersion": "apps/v1",
+            "kind": "Deployment",
+            "name": deploy.name,
+            "uid": deploy.uid,
+            "controller": True,
+            "blockOwnerDeletion": True,
+        }
+        key = (va.namespace, va.name)
+        stored = self.vas.get(key)
+        refs = [r for r in (stored.ownerReferences if stored else []) if r.get("uid") != deploy.uid]
+        refs.append(ref)
+        if stored is not None:
+            stored.ownerReferences = refs
+        va.ownerReferences = refs
+
+
+DEFAULT_SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
+class HttpKube:
+    """REST client against the API server (in-cluster by default)."""
+
+    def __init__(
+        self,
+        base_url: Optional[str] = None,
+        token: Optional[str] = None,
+        ca_cert: Optional[str] = None,
+        verify: bool | str = True,
+        timeout: float = 15.0,
+    ):
+        import httpx
+
+        if base_url is None:
+            host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes.default.svc")
+            port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            base_url = f"https://{host}:{port}"
+        if token is None:
+            token_path = os.path.join(DEFAULT_SA_DIR, "token")
+            if os.path.exists(token_path):
+                with open(token_path) as f:
+                    token = f.read().strip()
+        if ca_cert is None:
+            ca_path = os.path.join(DEFAULT_SA_DIR, "ca.crt")
+            if os.path.exists(ca_path):
+                ca_cert = ca_path
+        headers = {"Content-Type": "application/json"}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._client = httpx.Client(
+            base_url=base_url,
+            headers=headers,
+            verify=ca_cert if ca_cert else verify,
+            timeout=timeout,
+        )
+
+    # -- helpers -------------------------------------------------------
+    def _get(self, path: str) -> Optional[dict]:
+        r = self._client.get(path)
+        if r.status_code == 404:
+            return None
+        r.raise_for_status()
+        return r.json()
+
+    # -- KubeClient ----------------------------------------------------
+    def list_variantautoscalings(self) -> list[api.VariantAutoscaling]:
+        doc = self._get(f"/apis/{api.GROUP}/{api.VERSION}/{api.PLURAL}")
+        if doc is None:
+            return []
+        return [api.va_from_json(item) for item in doc.get("items", [])]
+
+    def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]:
+        doc = self._get(f"/api/v1/namespaces/{namespace}/configmaps/{name}")
+        return None if doc is None else dict(doc.get("data", {}) or {})
+
+    def get_deployment(self, namespace: str, name: str) -> Optional[Deployment]:
+        doc = self._get(f"/apis/apps/v1/namespaces/{namespace}/deployments/{name}")
+        if doc is None:
+            return None
+        meta = doc.get("metadata", {}) or {}
+        spec = doc.get("spec", {}) or {}
+        status = doc.get("status", {}) or {}
+        return Deployment(
+            name=meta.get("name", name),
+            namespace=meta.get("namespace", namespace),
+            replicas=int(spec.get("replicas", 1) or 0),
+            status_replicas=int(status.get("replicas", 0) or 0),
+            uid=meta.get("uid", ""),
+            labels=dict(meta.get("labels", {}) or {}),
+        )
+
+    def update_va_status(self, va: api.VariantAutoscaling) -> None:
+        path = (
+            f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/"
+            f"{api.PLURAL}/{va.name}/status"
+        )
+        body = {
+            "apiVersion": api.API_VERSION,
+            "kind": api.KIND,
+            "metadata": {"name": va.name, "namespace": va.namespace},
+            "status": api.va_to_json(va)["status"],
+        }
+        r = self._client.patch(
+            path, json=body, headers={"Content-Type": "application/merge-patch+json"}
+        )
+        r.raise_for_status()
+
+    def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
+        path = f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/{api.PLURAL}/{va.name}"
+        ref = {
+            "apiVersion": "apps/v1",
+            "kind": "Deployment",
+            "name": deploy.name,
+            "uid": deploy.uid,
+            "controller": True,
+            "blockOwnerDeletion": True,
+        }
+        refs = [r for r in va.ownerReferences if r.get("uid") != deploy.uid]
+        refs.append(ref)
+        body = {"metadata": {"ownerReferences": refs}}
+        r = self._client.patch(
+            path, json=body, headers={"Content-Type": "application/merge-patch+json"}
+        )
+        r.raise_for_status()
+        va.ownerReferences = refs

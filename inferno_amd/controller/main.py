@@ -1,0 +1,144 @@
+"""Controller process entrypoint.
+
+Equivalent of cmd/main.go:62-275: flag/env parsing, JSON logging, metrics
+endpoint, health probes, lease-based leader election, GPU device init/health
+probe (MI355X addition), then the reconcile loop at GLOBAL_OPT_INTERVAL.
+
+Run: python -m inferno_amd.controller.main [--metrics-port 8443]
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import signal
+import socket
+import threading
+import time
+
+from ..utils.logging import init_logger
+from . import collector
+from .k8s import HttpKube
+from .leader import LeaderElector
+from .metrics import init_metrics
+from .reconciler import CONFIGMAP_NAMESPACE, Reconciler, WVA_CONFIG_CM
+
+
+def gpu_health_probe() -> dict:
+    """MI355X device probe used by readyz (HIP addition over the reference)."""
+    try:
+        import torch
+
+        if not torch.cuda.is_available():
+            return {"gpu": False, "reason": "no HIP device"}
+        from ..ops.sweep import load_library
+
+        load_library(allow_build=False)
+        return {"gpu": True, "device": torch.cuda.get_device_name(0)}
+    except Exception as e:  # noqa: BLE001
+        return {"gpu": False, "reason": str(e)}
+
+
+def serve_probes(port: int, state: dict) -> None:
+    """healthz/readyz + prometheus metrics on one HTTP port."""
+    from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+    from prometheus_client import generate_latest
+
+    class Handler(BaseHTTPRequestHandler):
+        def do_GET(self):  # noqa: N802
+            if self.path == "/healthz":
+                self._ok(b"ok")
+            elif self.path == "/readyz":
+                if state.get("ready"):
+                    self._ok(b"ok")
+                else:
+                    self.send_response(503)
+                    self.end_headers()
+            elif self.path == "/metrics":
+                body = generate_latest()
+                self.send_response(200)
+                self.send_header("Content-Type", "text/plain; version=0.0.4")
+                self.end_headers()
+                self.wfile.write(body)
+            else:
+                self.send_response(404)
+                self.end_headers()
+
+        def _ok(self, body: bytes):
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):  # silence
+            pass
+
+    server = ThreadingHTTPServer(("0.0.0.0", port), Handler)
+    threading.Thread(target=server.serve_forever, daemon=True).start()
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--metrics-port", type=int, default=8443)
+    p.add_argument("--leader-elect", action="store_true", default=True)
+    p.add_argument("--no-leader-elect", dest="leader_elect", action="store_false")
+    p.add_argument("--configmap-namespace", default=CONFIGMAP_NAMESPACE)
+    p.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
+    args = p.parse_args()
+
+    logger = init_logger()
+    state = {"ready": False}
+    serve_probes(args.metrics_port, state)
+
+    kube = HttpKube()
+    prom_url = os.environ.get("PROMETHEUS_BASE_URL", "")
+    if not prom_url:
+        cm = kube.get_configmap(args.configmap_namespace, WVA_CONFIG_CM) or {}
+        prom_url = cm.get("PROMETHEUS_BASE_URL", "")
+    prom = collector.PrometheusClient(
+        prom_url,
+        token=os.environ.get("PROMETHEUS_BEARER_TOKEN") or None,
+        ca_cert=os.environ.get("PROMETHEUS_CA_CERT_PATH") or None,
+        insecure_skip_verify=os.environ.get("PROMETHEUS_TLS_INSECURE_SKIP_VERIFY") == "true",
+        allow_http=os.environ.get("PROMETHEUS_ALLOW_HTTP") == "true",
+    )
+
+    emitter = init_metrics()
+    reconciler = Reconciler(
+        kube, prom, emitter, backend=args.backend,
+        configmap_namespace=args.configmap_namespace,
+    )
+
+    logger.info("starting controller", extra={"kv": gpu_health_probe()})
+
+    elector = None
+    if args.leader_elect:
+        identity = f"{socket.gethostname()}_{os.getpid()}"
+        # LeaderElectionID mirrors the reference (cmd/main.go:207)
+        elector = LeaderElector(kube, "72dd1cf1.llm-d.ai", args.configmap_namespace, identity)
+
+    stop = threading.Event()
+    signal.signal(signal.SIGTERM, lambda *_: stop.set())
+    signal.signal(signal.SIGINT, lambda *_: stop.set())
+
+    state["ready"] = True
+    while not stop.is_set():
+        if elector is not None and not elector.try_acquire():
+            time.sleep(2.0)
+            continue
+        result = reconciler.reconcile()
+        logger.info(
+            "reconcile complete",
+            extra={
+                "kv": {
+                    "processed": result.processed,
+                    "backend": result.solver_backend,
+                    "duration_s": round(result.duration_seconds, 4),
+                    "errors": result.errors[:5],
+                }
+            },
+        )
+        stop.wait(result.requeue_after)
+
+
+if __name__ == "__main__":
+    main()

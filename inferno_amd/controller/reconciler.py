@@ -1,0 +1,313 @@
+"""VariantAutoscaling reconciler — the per-tick global optimization loop.
+
+Mirrors internal/controller/variantautoscaling_controller.go:86-407:
+read the three ConfigMaps, list active VAs, build SystemData via the
+adapters, validate + collect Prometheus metrics per VA (continue-on-error so
+one bad variant never blocks the fleet), run the batched analyzer+solver
+(HIP sweep on MI355X; CPU golden fallback raises the ``SolverDegraded``
+reason), then write DesiredOptimizedAlloc + conditions into each VA status
+and emit the inferno_* gauges that HPA/KEDA consume. Actuation is
+signal-only: the controller never patches Deployment replicas
+(actuator.go:50-84).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from datetime import datetime, timezone
+from typing import Optional
+
+from ..api import v1alpha1 as api
+from ..core.system import System
+from ..engine import SweepEngine
+from ..parallel import ShardedSolver
+from . import adapters, collector
+from .k8s import Deployment, KubeClient
+from .metrics import MetricsEmitter
+
+CONFIGMAP_NAMESPACE = "workload-variant-autoscaler-system"
+ACCELERATOR_CM = "accelerator-unit-costs"
+SERVICE_CLASS_CM = "service-classes-config"
+WVA_CONFIG_CM = "workload-variant-autoscaler-variantautoscaling-config"
+DEFAULT_INTERVAL_SECONDS = 60.0
+
+
+def _now_iso() -> str:
+    return datetime.now(timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
+
+
+def parse_go_duration(s: str) -> float:
+    """Parse a Go-style duration string ("60s", "1m30s", "500ms") to seconds."""
+    import re
+
+    if not s:
+        raise ValueError("empty duration")
+    units = {"h": 3600.0, "m": 60.0, "s": 1.0, "ms": 1e-3, "us": 1e-6, "ns": 1e-9}
+    total = 0.0
+    matched = False
+    for num, unit in re.findall(r"([0-9]*\.?[0-9]+)(h|ms|us|ns|m|s)", s):
+        total += float(num) * units[unit]
+        matched = True
+    if not matched:
+        raise ValueError(f"invalid duration {s!r}")
+    return total
+
+
+@dataclass
+class ReconcileResult:
+    requeue_after: float = DEFAULT_INTERVAL_SECONDS
+    processed: int = 0
+    solver_backend: str = ""
+    degraded: bool = False
+    errors: list[str] = field(default_factory=list)
+    duration_seconds: float = 0.0
+
+
+class Actuator:
+    """Signal-only actuation: gauges from real Deployment replica counts.
+
+    Ref internal/actuator/actuator.go:28-84.
+    """
+
+    def __init__(self, kube: KubeClient, emitter: MetricsEmitter):
+        self.kube = kube
+        self.emitter = emitter
+
+    def current_deployment_replicas(self, va: api.VariantAutoscaling) -> int:
+        deploy = self.kube.get_deployment(va.namespace, va.name)
+        if deploy is None:
+            # fallback to the (possibly stale) VA status
+            return va.status.currentAlloc.numReplicas
+        if deploy.status_replicas >= 0:
+            return deploy.status_replicas
+        if deploy.replicas is not None:
+            return deploy.replicas
+        return 1
+
+    def emit_metrics(self, va: api.VariantAutoscaling) -> None:
+        if va.status.desiredOptimizedAlloc.numReplicas < 0:
+            return
+        current = self.current_deployment_replicas(va)
+        self.emitter.emit_replica_metrics(
+            va.name,
+            va.namespace,
+            current,
+            va.status.desiredOptimizedAlloc.numReplicas,
+            va.status.desiredOptimizedAlloc.accelerator,
+        )
+
+
+class Reconciler:
+    def __init__(
+        self,
+        kube: KubeClient,
+        prom: collector.PromAPI,
+        emitter: MetricsEmitter,
+        backend: str = "auto",
+        configmap_namespace: str = CONFIGMAP_NAMESPACE,
+        scale_to_zero: Optional[bool] = None,
+    ):
+        self.kube = kube
+        self.prom = prom
+        self.emitter = emitter
+        self.actuator = Actuator(kube, emitter)
+        self.engine = SweepEngine(backend=backend)
+        self.solver = ShardedSolver(self.engine)
+        self.configmap_namespace = configmap_namespace
+        self.scale_to_zero = scale_to_zero
+        self.last_result: Optional[ReconcileResult] = None
+
+    # ------------------------------------------------------------------
+    def read_interval(self) -> float:
+        cm = self.kube.get_configmap(self.configmap_namespace, WVA_CONFIG_CM) or {}
+        interval = cm.get("GLOBAL_OPT_INTERVAL", "")
+        if interval:
+            try:
+                return parse_go_duration(interval)
+            except ValueError:
+                pass
+        return DEFAULT_INTERVAL_SECONDS
+
+    def _read_accelerator_cm(self) -> Optional[dict[str, str]]:
+        return self.kube.get_configmap(self.configmap_namespace, ACCELERATOR_CM)
+
+    def _read_service_class_cm(self) -> Optional[dict[str, str]]:
+        return self.kube.get_configmap(self.configmap_namespace, SERVICE_CLASS_CM)
+
+    # ------------------------------------------------------------------
+    def reconcile(self) -> ReconcileResult:
+        t_start = time.perf_counter()
+        result = ReconcileResult(requeue_after=self.read_interval())
+
+        accelerator_cm = self._read_accelerator_cm()
+        if accelerator_cm is None:
+            result.errors.append("unable to read accelerator configMap")
+            return self._finish(result, t_start)
+        service_class_cm = self._read_service_class_cm()
+        if service_class_cm is None:
+            result.errors.append("unable to read serviceclass configMap")
+            return self._finish(result, t_start)
+
+        vas = [va for va in self.kube.list_variantautoscalings() if not va.deletionTimestamp]
+        if not vas:
+            return self._finish(result, t_start)
+
+        spec = adapters.create_system_data(accelerator_cm, service_class_cm)
+        import json
+
+        acc_costs: dict[str, float] = {}
+        for key, val in accelerator_cm.items():
+            try:
+                acc_costs[key] = float(json.loads(val)["cost"])
+            except (json.JSONDecodeError, KeyError, TypeError, ValueError):
+                continue
+
+        # ---- prepare phase (per-VA, continue on error) -------------------
+        update_list: list[api.VariantAutoscaling] = []
+        for va in vas:
+            model_name = va.spec.modelID
+            if not model_name:
+                continue
+            try:
+                _, class_name = adapters.find_model_slo(service_class_cm, model_name)
+            except adapters.AdapterError as e:
+                result.errors.append(f"{va.name}: {e}")
+                continue
+            for profile in va.spec.modelProfile.accelerators:
+                try:
+                    adapters.add_model_accelerator_profile(spec, model_name, profile)
+                except adapters.AdapterError:
+                    continue
+            acc_name = va.labels.get(api.ACCELERATOR_LABEL, "")
+            if acc_name not in acc_costs:
+                result.errors.append(f"{va.name}: missing accelerator cost for {acc_name!r}")
+                continue
+            deploy = self.kube.get_deployment(va.namespace, va.name)
+            if deploy is None:
+                result.errors.append(f"{va.name}: deployment not found")
+                continue
+            if not any(r.get("uid") == deploy.uid for r in va.ownerReferences):
+                self.kube.set_owner_reference(va, deploy)
+
+            validation = collector.validate_metrics_availability(
+                self.prom, model_name, deploy.namespace
+            )
+            if validation.available:
+                api.set_condition(
+                    va,
+                    api.TYPE_METRICS_AVAILABLE,
+                    "True",
+                    validation.reason,
+                    validation.message,
+                )
+            else:
+                # metrics unavailable: log and skip (ref controller.go:305-316)
+                result.errors.append(f"{va.name}: metrics unavailable ({validation.reason})")
+                continue
+
+            try:
+                current_alloc = collector.add_metrics_to_opt_status(
+                    va, deploy.namespace, deploy.replicas, acc_costs[acc_name], self.prom
+                )
+            except Exception as e:
+                result.errors.append(f"{va.name}: metric collection failed: {e}")
+                continue
+            va.status.currentAlloc = current_alloc
+
+            try:
+                adapters.add_server_info(spec, va, class_name, self.scale_to_zero)
+            except adapters.AdapterError as e:
+                result.errors.append(f"{va.name}: bad server data: {e}")
+                continue
+            update_list.append(va)
+
+        if not update_list:
+            return self._finish(result, t_start)
+
+        # ---- analyze + optimize (batched sweep) --------------------------
+        system, opt_spec = System.from_spec(spec)
+        for acc in system.accelerators.values():
+            acc.calculate()
+        try:
+            shard_result = self.solver.solve(system, opt_spec)
+            solution = shard_result.solution
+            result.solver_backend = self.engine.backend
+        except Exception as e:
+            if self.engine.backend == "gpu":
+                # GPU failure -> CPU reference fallback with SolverDegraded
+                # (SURVEY.md section 5 failure-detection plan)
+                self.engine = SweepEngine(backend="cpu")
+                self.solver = ShardedSolver(self.engine)
+                try:
+                    shard_result = self.solver.solve(system, opt_spec)
+                    solution = shard_result.solution
+                    result.solver_backend = "cpu"
+                    result.degraded = True
+                except Exception as e2:
+                    return self._fail_all(update_list, result, t_start, e2)
+            else:
+                return self._fail_all(update_list, result, t_start, e)
+
+        if not solution:
+            return self._fail_all(
+                update_list, result, t_start,
+                RuntimeError("no feasible allocations found for all variants"),
+            )
+
+        # ---- apply phase -------------------------------------------------
+        now = _now_iso()
+        for va in update_list:
+            try:
+                optimized = adapters.create_optimized_alloc(va.name, va.namespace, solution, now)
+            except adapters.AdapterError:
+                continue
+            va.status.desiredOptimizedAlloc = optimized
+            va.status.actuation.applied = False
+            reason = api.REASON_OPTIMIZATION_SUCCEEDED
+            message = (
+                f"Optimization completed: {optimized.numReplicas} replicas on "
+                f"{optimized.accelerator}"
+            )
+            if result.degraded:
+                reason = api.REASON_SOLVER_DEGRADED
+                message += " (GPU solver degraded to CPU reference)"
+            api.set_condition(va, api.TYPE_OPTIMIZATION_READY, "True", reason, message)
+            try:
+                self.actuator.emit_metrics(va)
+                va.status.actuation.applied = True
+            except Exception as e:
+                result.errors.append(f"{va.name}: metric emission failed: {e}")
+            try:
+                self.kube.update_va_status(va)
+            except Exception as e:
+                result.errors.append(f"{va.name}: status update failed: {e}")
+                continue
+            result.processed += 1
+
+        return self._finish(result, t_start)
+
+    # ------------------------------------------------------------------
+    def _fail_all(self, update_list, result, t_start, err) -> ReconcileResult:
+        for va in update_list:
+            api.set_condition(
+                va,
+                api.TYPE_OPTIMIZATION_READY,
+                "False",
+                api.REASON_OPTIMIZATION_FAILED,
+                f"Optimization failed: {err}",
+            )
+            try:
+                self.kube.update_va_status(va)
+            except Exception:
+                pass
+        result.errors.append(f"optimization failed: {err}")
+        return self._finish(result, t_start)
+
+    def _finish(self, result: ReconcileResult, t_start: float) -> ReconcileResult:
+        result.duration_seconds = time.perf_counter() - t_start
+        try:
+            self.emitter.reconcile_latency.observe(result.duration_seconds)
+        except Exception:
+            pass
+        self.last_result = result
+        return result
