@@ -1,0 +1,164 @@
+"""MI355X perf-model derivation sanity: TP scaling directions, KV-cache
+sizing against 288 GB HBM3E, xGMI all-reduce cost model, CR wire format,
+and end-to-end use of a derived fleet in the sweep engine."""
+import pytest
+
+from inferno_amd.perfmodel import (
+    LLAMA_8B,
+    LLAMA_70B,
+    MI300X,
+    MI355X,
+    allreduce_ms,
+    derive_perf_data,
+    derive_profile,
+    tp_variant_name,
+)
+
+
+class TestAllReduce:
+    def test_tp1_free(self):
+        assert allreduce_ms(1 << 20, 1, MI355X) == 0.0
+
+    def test_grows_with_tp_latency(self):
+        # small message: latency-dominated, grows with hop count
+        a2 = allreduce_ms(8192 * 2, 2, MI355X)
+        a8 = allreduce_ms(8192 * 2, 8, MI355X)
+        assert a8 > a2 > 0
+
+    def test_bandwidth_term(self):
+        # large message: ~2(G-1)/G * M / link_bw
+        M = 1e9
+        t4 = allreduce_ms(M, 4, MI355X)
+        expect = (2 * 3 / 4 * M / (153e9) + 3 * 2e-6) * 1e3
+        assert t4 == pytest.approx(expect, rel=1e-6)
+
+
+class TestDeriveProfile:
+    def test_decode_is_hbm_bound(self):
+        p = derive_profile(LLAMA_8B, MI355X, tp=1)
+        # 16 GB of bf16 weights over ~6.3 TB/s ~ 2.5 ms + overhead
+        assert 2.0 < p.alpha < 5.0
+        assert p.beta > 0
+
+    def test_alpha_improves_with_tp(self):
+        p1 = derive_profile(LLAMA_70B, MI355X, tp=1)
+        p4 = derive_profile(LLAMA_70B, MI355X, tp=4)
+        p8 = derive_profile(LLAMA_70B, MI355X, tp=8)
+        # weight streaming shards with TP (communication partly offsets)
+        assert p8.alpha < p4.alpha < p1.alpha
+
+    def test_gamma_comm_floor_grows_with_tp(self):
+        p1 = derive_profile(LLAMA_70B, MI355X, tp=1)
+        p8 = derive_profile(LLAMA_70B, MI355X, tp=8)
+        assert p8.gamma > p1.gamma  # all-reduce latency floor
+
+    def test_delta_scales_inverse_tp(self):
+        p1 = derive_profile(LLAMA_8B, MI355X, tp=1)
+        p2 = derive_profile(LLAMA_8B, MI355X, tp=2)
+        assert p2.delta == pytest.approx(p1.delta / 2, rel=1e-6)
+
+    def test_max_batch_kv_sizing(self):
+        p = derive_profile(LLAMA_8B, MI355X, tp=1, at_tokens=1024)
+        # usable = 288e9*0.9 - 16e9; kv/req = 2*32*8*128*2*1024
+        usable = 288e9 * 0.9 - 16e9
+        kv_per_req = 2 * 32 * 8 * 128 * 2 * 1024
+        assert p.max_batch_size == int(usable // kv_per_req)
+
+    def test_max_batch_grows_with_tp(self):
+        p1 = derive_profile(LLAMA_70B, MI355X, tp=1)
+        p8 = derive_profile(LLAMA_70B, MI355X, tp=8)
+        assert p8.max_batch_size > p1.max_batch_size
+
+    def test_mi355x_beats_mi300x(self):
+        a = derive_profile(LLAMA_8B, MI355X, tp=1)
+        b = derive_profile(LLAMA_8B, MI300X, tp=1)
+        assert a.alpha < b.alpha
+        assert a.delta < b.delta
+        assert a.max_batch_size > b.max_batch_size
+
+    def test_oversize_model_returns_none(self):
+        from inferno_amd.perfmodel import LlmSpec
+
+        huge = LlmSpec("huge", 400.0, 120, 16384, 128, 16)
+        assert derive_profile(huge, MI300X, tp=1) is None
+        assert derive_profile(huge, MI355X, tp=8) is not None
+
+
+class TestWireFormat:
+    def test_perf_parms_strings(self):
+        p = derive_profile(LLAMA_8B, MI355X, tp=2)
+        parms = p.perf_parms()
+        assert set(parms.decodeParms) == {"alpha", "beta"}
+        assert set(parms.prefillParms) == {"gamma", "delta"}
+        float(parms.decodeParms["alpha"])  # parseable like the CR adapter does
+
+    def test_variant_rows(self):
+        rows = derive_perf_data(LLAMA_70B, gpus=[MI355X], tps=(1, 2, 4, 8))
+        names = {r.acc for r in rows}
+        assert names == {"MI355X", "MI355X-TP2", "MI355X-TP4", "MI355X-TP8"}
+        for r in rows:
+            assert r.accCount == (1 if r.acc == "MI355X" else int(r.acc.split("TP")[1]))
+            assert r.maxBatchSize >= 1 and r.atTokens == 1024
+
+    def test_variant_name(self):
+        assert tp_variant_name(MI355X, 1) == "MI355X"
+        assert tp_variant_name(MI355X, 4) == "MI355X-TP4"
+
+
+class TestEndToEnd:
+    def test_derived_fleet_solves(self):
+        """TP variants of a 70B model on MI355X compete in the sweep: the
+        solver picks a feasible, SLO-meeting, min-penalty variant."""
+        from inferno_amd.config import (
+            AcceleratorCount,
+            AllocationData,
+            ModelTarget,
+            OptimizerSpec,
+            ServerLoadSpec,
+            ServerSpec,
+            ServiceClassSpec,
+            SystemSpec,
+        )
+        from inferno_amd.core import System
+        from inferno_amd.engine import SweepEngine
+        from inferno_amd.perfmodel import accelerator_spec
+
+        rows = derive_perf_data(LLAMA_70B, gpus=[MI355X], tps=(1, 2, 4, 8))
+        accs = [accelerator_spec(MI355X, tp) for tp in (1, 2, 4, 8)]
+        spec = SystemSpec(
+            accelerators=accs,
+            models=rows,
+            serviceClasses=[
+                ServiceClassSpec(
+                    name="Premium",
+                    priority=1,
+                    modelTargets=[
+                        ModelTarget(model=LLAMA_70B.name, slo_itl=40.0, slo_ttft=2000.0)
+                    ],
+                )
+            ],
+            servers=[
+                ServerSpec(
+                    name="llama70b:prod",
+                    klass="Premium",
+                    model=LLAMA_70B.name,
+                    minNumReplicas=1,
+                    currentAlloc=AllocationData(
+                        accelerator="MI355X-TP4",
+                        numReplicas=1,
+                        cost=95.0 * 4,
+                        load=ServerLoadSpec(
+                            arrivalRate=600.0, avgInTokens=512, avgOutTokens=256
+                        ),
+                    ),
+                )
+            ],
+            optimizer=OptimizerSpec(unlimited=True),
+        )
+        system, opt = System.from_spec(spec)
+        SweepEngine(backend="cpu").solve(system, opt)
+        alloc = system.servers["llama70b:prod"].allocation
+        assert alloc is not None
+        assert alloc.accelerator in {"MI355X", "MI355X-TP2", "MI355X-TP4", "MI355X-TP8"}
+        # the winning variant meets the ITL SLO
+        assert alloc.itl <= 40.0 * 1.01
