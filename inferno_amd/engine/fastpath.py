@@ -196,65 +196,178 @@ class FastSweep:
             return self._reconcile_gpu()
         return self._reconcile_cpu()
 
+    # GPU persistent-state reconcile: static SoA uploaded once; per step the
+    # dynamic fields are packed into TWO pinned staging buffers (one int32,
+    # one fp32) -> two H2D copies; bucket ids cached on the batch_n bytes;
+    # outputs preallocated; winner records gathered ON-DEVICE and downloaded
+    # in a single D2H copy.
+    _DYN_INT = ("in_tok", "out_tok", "batch_n", "perf_max_batch", "cur_replicas", "flags")
+    _STAT_F32 = ("alpha", "beta", "gamma", "delta", "t_itl", "t_ttft", "t_tps", "acc_cost")
+
+    def _init_gpu_state(self) -> None:
+        import torch
+
+        from ..ops.sweep import load_library
+
+        load_library(allow_build=False)
+        dev = self.device
+        n = self.n_cells
+        st: dict = {}
+        for k in self._STAT_F32:
+            st[k] = torch.from_numpy(np.ascontiguousarray(self._stat[k])).to(dev)
+        st["min_replicas"] = torch.from_numpy(
+            np.ascontiguousarray(self._stat["min_replicas"], dtype=np.int32)
+        ).to(dev)
+        st["pin_i"] = torch.empty((len(self._DYN_INT), n), dtype=torch.int32,
+                                  pin_memory=True)
+        st["pin_f"] = torch.empty((2, n), dtype=torch.float32, pin_memory=True)
+        st["dev_i"] = torch.empty((len(self._DYN_INT), n), dtype=torch.int32, device=dev)
+        st["dev_f"] = torch.empty((2, n), dtype=torch.float32, device=dev)
+        st["seg"] = torch.from_numpy(self.seg_start).to(dev)
+        st["cell_acc"] = torch.from_numpy(self.cell_acc_idx).to(dev)
+        # outputs
+        st["feasible"] = torch.zeros(n, dtype=torch.uint8, device=dev)
+        st["zero_empty"] = torch.zeros(n, dtype=torch.uint8, device=dev)
+        st["num_replicas"] = torch.zeros(n, dtype=torch.int32, device=dev)
+        st["batch"] = torch.zeros(n, dtype=torch.int32, device=dev)
+        for k in ("cost", "value", "itl", "ttft", "rho", "max_rate"):
+            st[k] = torch.zeros(n, dtype=torch.float32, device=dev)
+        n_srv = len(self.server_names)
+        st["winner"] = torch.full((n_srv,), -1, dtype=torch.int32, device=dev)
+        st["gather_f"] = torch.empty((6, n_srv), dtype=torch.float32, device=dev)
+        st["gather_i"] = torch.empty((4, n_srv), dtype=torch.int32, device=dev)
+        st["pin_out_f"] = torch.empty((6, n_srv), dtype=torch.float32, pin_memory=True)
+        st["pin_out_i"] = torch.empty((4, n_srv), dtype=torch.int32, pin_memory=True)
+        st["bucket_key"] = None
+        st["buckets"] = []
+        st["side_streams"] = []
+        self._gpu = st
+
+    def _buckets_for(self, batch_n: np.ndarray):
+        import torch
+
+        from ..ops.sweep import N_MED, N_SMALL
+
+        st = self._gpu
+        key = batch_n.tobytes()
+        if st["bucket_key"] == key:
+            return st["buckets"]
+        buckets = []
+        small = batch_n <= N_SMALL
+        med = (batch_n > N_SMALL) & (batch_n <= N_MED)
+        large = batch_n > N_MED
+        for nt, mask in ((64, small), (256, med), (1024, large)):
+            idx = np.nonzero(mask)[0]
+            if len(idx):
+                ids = None
+                if len(idx) != len(batch_n):
+                    ids = torch.from_numpy(idx.astype(np.int32)).to(self.device)
+                buckets.append((nt, ids, int(batch_n[mask].max()), len(idx)))
+        st["bucket_key"] = key
+        st["buckets"] = buckets
+        while len(st["side_streams"]) < max(len(buckets) - 1, 0):
+            st["side_streams"].append(torch.cuda.Stream())
+        return buckets
+
     def _reconcile_gpu(self) -> WinnerRecord:
-        import os
-        import time
+        import ctypes
 
         import torch
 
-        from ..ops.sweep import run_argmin, run_sweep
+        from ..ops.sweep import HipKernelError, load_library
 
         n_srv = len(self.server_names)
         if self.n_cells == 0:
             return _empty_winner(n_srv)
-        trace = os.environ.get("INFERNO_TIMING", "") == "1"
-        if trace:
-            torch.cuda.synchronize()
-            t0 = time.perf_counter()
-        arrays = self.cell_arrays()
-        if trace:
-            t1 = time.perf_counter()
-        out = run_sweep(arrays, device=self.device)
-        seg = torch.from_numpy(self.seg_start).to(self.device)
-        winner = run_argmin(out.value, out.feasible, seg)
-        if trace:
-            torch.cuda.synchronize()
-            t2 = time.perf_counter()
-            print(
-                f"[timing] refresh={1e3*(t1-t0):.2f}ms sweep+argmin={1e3*(t2-t1):.2f}ms",
-                flush=True,
-            )
-        # single D2H sync for everything needed to materialize winners
-        w = winner.cpu().numpy()
-        stack = torch.stack(
-            [
-                out.cost,
-                out.value,
-                out.itl,
-                out.ttft,
-                out.rho,
-                out.max_rate,
-            ]
-        ).cpu().numpy()
-        reps = out.num_replicas.cpu().numpy()
-        batch = out.batch.cpu().numpy()
-        zero_empty = out.zero_empty.cpu().numpy()
+        if not hasattr(self, "_gpu"):
+            self._init_gpu_state()
+        st = self._gpu
+        lib = load_library(allow_build=False)
+        arrs = self._refresh_dynamic()
 
+        # pack + upload dynamics (2 copies)
+        pin_i, pin_f = st["pin_i"], st["pin_f"]
+        for j, k in enumerate(self._DYN_INT):
+            pin_i[j] = torch.from_numpy(np.ascontiguousarray(arrs[k], dtype=np.int32))
+        pin_f[0] = torch.from_numpy(np.ascontiguousarray(arrs["arrival_rate"], dtype=np.float32))
+        pin_f[1] = torch.from_numpy(np.ascontiguousarray(arrs["cur_cost"], dtype=np.float32))
+        st["dev_i"].copy_(pin_i, non_blocking=True)
+        st["dev_f"].copy_(pin_f, non_blocking=True)
+
+        di, df = st["dev_i"], st["dev_f"]
+
+        def p(t):
+            return ctypes.c_void_p(t.data_ptr())
+
+        main_stream = torch.cuda.current_stream()
+        buckets = self._buckets_for(arrs["batch_n"])
+        for i, (nt, ids, bmax, n_blocks) in enumerate(buckets):
+            if i == 0:
+                cur = main_stream
+            else:
+                cur = st["side_streams"][i - 1]
+                cur.wait_stream(main_stream)
+            rc = lib.wva_sweep_launch_bucket(
+                ctypes.c_int(n_blocks),
+                ctypes.c_int(max(bmax, 1)),
+                ctypes.c_int(nt),
+                p(ids) if ids is not None else None,
+                ctypes.c_void_p(cur.cuda_stream),
+                p(di[0]), p(di[1]), p(di[2]), p(st["min_replicas"]), p(di[3]),
+                p(di[4]), p(di[5]),
+                p(st["alpha"]), p(st["beta"]), p(st["gamma"]), p(st["delta"]),
+                p(df[0]),
+                p(st["t_itl"]), p(st["t_ttft"]), p(st["t_tps"]), p(st["acc_cost"]),
+                p(df[1]),
+                p(st["feasible"]), p(st["zero_empty"]), p(st["num_replicas"]),
+                p(st["batch"]), p(st["cost"]), p(st["value"]), p(st["itl"]),
+                p(st["ttft"]), p(st["rho"]), p(st["max_rate"]),
+            )
+            if rc != 0:
+                raise HipKernelError(f"wva_sweep_launch_bucket(nt={nt}) failed: {rc}")
+        for s in st["side_streams"][: max(len(buckets) - 1, 0)]:
+            main_stream.wait_stream(s)
+
+        rc = lib.wva_argmin_launch(
+            ctypes.c_int(n_srv),
+            ctypes.c_void_p(main_stream.cuda_stream),
+            p(st["value"]), p(st["feasible"]), p(st["seg"]), p(st["winner"]),
+        )
+        if rc != 0:
+            raise HipKernelError(f"wva_argmin_launch failed: {rc}")
+
+        # gather winner fields on-device, single pinned D2H
+        w = st["winner"]
         has = w >= 0
-        wc = np.where(has, w, 0)
-        acc_idx = np.where(
-            has, np.where(zero_empty[wc] > 0, -2, self.cell_acc_idx[wc]), -1
-        ).astype(np.int32)
+        wc = torch.where(has, w, torch.zeros_like(w)).long()
+        gf, gi = st["gather_f"], st["gather_i"]
+        zero_f = torch.zeros((), dtype=torch.float32, device=w.device)
+        for j, k in enumerate(("cost", "value", "itl", "ttft", "rho", "max_rate")):
+            gf[j] = torch.where(has, st[k][wc], zero_f)
+        acc = torch.where(
+            st["zero_empty"][wc] > 0,
+            torch.full_like(w, -2),
+            st["cell_acc"][wc],
+        )
+        gi[0] = torch.where(has, acc, torch.full_like(w, -1))
+        gi[1] = torch.where(has, st["num_replicas"][wc], torch.zeros_like(w))
+        gi[2] = torch.where(has, st["batch"][wc], torch.zeros_like(w))
+        gi[3] = w
+        st["pin_out_f"].copy_(gf, non_blocking=True)
+        st["pin_out_i"].copy_(gi, non_blocking=True)
+        torch.cuda.current_stream().synchronize()
+        of = st["pin_out_f"].numpy()
+        oi = st["pin_out_i"].numpy()
         return WinnerRecord(
-            acc_idx=acc_idx,
-            num_replicas=np.where(has, reps[wc], 0).astype(np.int32),
-            batch=np.where(has, batch[wc], 0).astype(np.int32),
-            cost=np.where(has, stack[0][wc], 0.0).astype(np.float32),
-            value=np.where(has, stack[1][wc], 0.0).astype(np.float32),
-            itl=np.where(has, stack[2][wc], 0.0).astype(np.float32),
-            ttft=np.where(has, stack[3][wc], 0.0).astype(np.float32),
-            rho=np.where(has, stack[4][wc], 0.0).astype(np.float32),
-            max_rate=np.where(has, stack[5][wc], 0.0).astype(np.float32),
+            acc_idx=oi[0].copy(),
+            num_replicas=oi[1].copy(),
+            batch=oi[2].copy(),
+            cost=of[0].copy(),
+            value=of[1].copy(),
+            itl=of[2].copy(),
+            ttft=of[3].copy(),
+            rho=of[4].copy(),
+            max_rate=of[5].copy(),
         )
 
     def _reconcile_cpu(self) -> WinnerRecord:
