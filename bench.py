@@ -60,7 +60,9 @@ def main():
     if use_dist:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29771")
-        backend = "nccl" if use_gpu else "gloo"
+        # INFERNO_DIST_BACKEND=gloo lets multi-rank GPU runs share one device
+        # (testing); production multi-GPU uses RCCL ("nccl" on ROCm)
+        backend = os.environ.get("INFERNO_DIST_BACKEND") or ("nccl" if use_gpu else "gloo")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
         if use_gpu:
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1))))
