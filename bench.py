@@ -65,7 +65,9 @@ def main():
         backend = os.environ.get("INFERNO_DIST_BACKEND") or ("nccl" if use_gpu else "gloo")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
         if use_gpu:
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank % max(torch.cuda.device_count(), 1))))
+            n_dev = max(torch.cuda.device_count(), 1)
+            local_rank = int(os.environ.get("LOCAL_RANK", rank))
+            torch.cuda.set_device(local_rank % n_dev)
 
     n_models = args.models_per_gpu * world
     spec = make_fleet_spec(n_models, seed=args.seed)
