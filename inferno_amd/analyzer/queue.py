@@ -363,6 +363,14 @@ class QueueAnalyzer:
         )
         return self.service_parms.decode.decode_time(eff)
 
+    def eval_serv_time(self, lam: float) -> float:
+        """Avg service time at lam (req/msec). Ref: pkg/analyzer/utils.go:75-81."""
+        return self.chain.solve(lam).avg_serv_time
+
+    def eval_waiting_time(self, lam: float) -> float:
+        """Avg waiting time at lam (req/msec). Ref: pkg/analyzer/utils.go:84-90."""
+        return self.chain.solve(lam).avg_wait_time
+
     # -- public API --------------------------------------------------------
 
     def analyze(self, request_rate: float) -> AnalysisMetrics:

@@ -4,6 +4,8 @@ from .allocation import (  # noqa: F401
     allocation_from_data,
     create_allocation,
     create_allocation_diff,
+    reallocate,
+    scale_allocation,
 )
 from .system import (  # noqa: F401
     Accelerator,

@@ -153,6 +153,33 @@ def _zero_load_allocation(server, model, acc, perf) -> Allocation:
     return alloc
 
 
+def scale_allocation(system, alloc: Allocation, server_name: str):
+    """Re-size this allocation's accelerator for the server's current load;
+    returns (new allocation, replica delta). Ref: allocation.go:166-190 Scale."""
+    server = system.servers.get(server_name)
+    if server is None or server.load is None:
+        return None, 0
+    if system.accelerators.get(alloc.accelerator) is None:
+        return None, 0
+    new = create_allocation(system, server_name, alloc.accelerator)
+    if new is None:
+        return None, 0
+    return new, new.num_replicas - alloc.num_replicas
+
+
+def reallocate(system, server_name: str):
+    """Min-value allocation across all accelerators; returns (allocation,
+    accelerator name) or (None, ""). Ref: allocation.go:192-207 ReAllocate."""
+    min_alloc: Optional[Allocation] = None
+    for g_name in sorted(system.accelerators):
+        alloc = create_allocation(system, server_name, g_name)
+        if alloc is not None and (min_alloc is None or alloc.value < min_alloc.value):
+            min_alloc = alloc
+    if min_alloc is None:
+        return None, ""
+    return min_alloc, min_alloc.accelerator
+
+
 def create_allocation(system, server_name: str, acc_name: str) -> Optional[Allocation]:
     """Size an accelerator for a server; None when infeasible.
 
