@@ -32,9 +32,43 @@ def _parse_args():
     p.add_argument("--models-per-gpu", type=int, default=512)
     p.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
     p.add_argument("--seed", type=int, default=1234)
-    p.add_argument("--baseline-cpu", action="store_true",
-                   help="also measure the scalar CPU reference on a small probe and report speedup")
+    p.add_argument(
+        "--preset",
+        choices=["config2", "config3", "config4", "config5"],
+        default=None,
+        help="BASELINE.json sweep configs: config2 = 8-model MI355X-only; "
+        "config3 = 64 models x 4 TP variants; config4 = 512-model fleet "
+        "across MI300X/MI325X/MI355X (default shape); config5 = 4096 models "
+        "x 8 variants (sharded)",
+    )
     return p.parse_args()
+
+
+def preset_fleet(preset, models_per_gpu, world, seed):
+    """Build the fleet spec for a BASELINE config preset."""
+    from inferno_amd.perfmodel import MI355X, accelerator_spec
+    from inferno_amd.utils.synthetic import AMD_ACCELERATORS, make_fleet_spec
+
+    if preset == "config2":
+        return make_fleet_spec(8 * world, seed=seed,
+                               accelerators=[AMD_ACCELERATORS[2]]), 8
+    if preset == "config3":
+        accs = [accelerator_spec(MI355X, tp) for tp in (1, 2, 4, 8)][:4]
+        return make_fleet_spec(64 * world, seed=seed, accelerators=accs), 64
+    if preset == "config5":
+        # 8 variants per model: the 3-SKU ladder + MI355X TP 2/4/8 + 2 spot tiers
+        from dataclasses import replace
+
+        accs = (
+            list(AMD_ACCELERATORS)
+            + [accelerator_spec(MI355X, tp) for tp in (2, 4, 8)]
+            + [
+                replace(AMD_ACCELERATORS[0], name="MI300X-spot", cost=45.0),
+                replace(AMD_ACCELERATORS[2], name="MI355X-spot", cost=70.0),
+            ]
+        )
+        return make_fleet_spec(4096 * world, seed=seed, accelerators=accs), 4096
+    return make_fleet_spec(models_per_gpu * world, seed=seed), models_per_gpu
 
 
 def main():
@@ -69,8 +103,9 @@ def main():
             local_rank = int(os.environ.get("LOCAL_RANK", rank))
             torch.cuda.set_device(local_rank % n_dev)
 
-    n_models = args.models_per_gpu * world
-    spec = make_fleet_spec(n_models, seed=args.seed)
+    spec, args.models_per_gpu = preset_fleet(
+        args.preset, args.models_per_gpu, world, args.seed
+    )
     system, opt_spec = System.from_spec(spec)
     for acc in system.accelerators.values():
         acc.calculate()
@@ -123,6 +158,7 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
         step_times.append((time.perf_counter() - t0) * 1000.0)
+        cells_per_step = r.local_stats.n_cells
     barrier_sync()
     elapsed = time.perf_counter() - t_start
 

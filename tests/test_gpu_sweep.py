@@ -192,3 +192,35 @@ class TestFastPathGpu:
             else:
                 # legitimate only on a near-tie of candidate values
                 assert g.cost == pytest.approx(c.cost, rel=0.25)
+
+
+class TestGreedyOnGpuSweep:
+    def test_greedy_limited_mode_uses_gpu_candidates(self):
+        cap = {"AMD-MI355X-288GB": 8, "AMD-MI325X-256GB": 8, "AMD-MI300X-192GB": 8}
+        a, opt = System.from_spec(make_spec(n_servers=6, seed=500, unlimited=False,
+                                            capacity=dict(cap)))
+        b, _ = System.from_spec(make_spec(n_servers=6, seed=500, unlimited=False,
+                                          capacity=dict(cap)))
+        SweepEngine(backend="gpu").solve(a, opt)
+        SweepEngine(backend="cpu").solve(b, opt)
+        used = {t: 0 for t in cap}
+        for srv in a.servers.values():
+            if srv.allocation is None:
+                continue
+            acc = a.accelerators[srv.allocation.accelerator]
+            model = a.models[srv.model_name]
+            used[acc.type] += (
+                srv.allocation.num_replicas
+                * model.get_num_instances(acc.name)
+                * acc.multiplicity
+            )
+        for t in cap:
+            assert used[t] <= cap[t]
+        # same allocation outcome as greedy over CPU candidates (tolerating
+        # near-tie ordering differences)
+        n_same = sum(
+            1
+            for n in a.servers
+            if (a.servers[n].allocation is None) == (b.servers[n].allocation is None)
+        )
+        assert n_same >= len(a.servers) - 1
