@@ -165,29 +165,68 @@ struct ChainOut {
   double in_servers;
 };
 
-// Solve the state-dependent chain at arrival rate lam. S = inclusive prefix
-// of log service rates in LDS; logsN = log s(N). All threads return
+// Chain geometry shared by the setup scan and every evaluation.
+// Each lane owns a CONTIGUOUS chunk of states n in [n0, n1]; within the
+// chunk, probabilities advance by the linear-space recurrence
+// w(n+1) = w(n) * lam / s(n+1) — one fp64 exp per SUB-state anchor instead
+// of one per state (the exp is the dominant per-state cost; a 32-state
+// sub-chunk product spans at most e^224 relative to its anchor, far inside
+// fp64 range, and anchors are pinned <= exp(0) by the anchor max).
+// LDS layout: inv_s_t is chunk-TRANSPOSED ([j*NT + lane] holds 1/s(n0+j))
+// so the per-step reads are conflict-free; S_anchor holds the log-prefix at
+// the anchor states only.
+#define WVA_SUB 32
+
+struct ChainGeom {
+  const double *inv_s_t;   // [chunk*NT] transposed reciprocal service rates
+  const double *S_anchor;  // [NT*ksub] log-prefix at anchors
+  double S_total;          // S[N]
+  int chunk;               // states per lane = ceil(N/NT)
+  int ksub;                // anchors per lane = ceil(chunk/SUB)
+};
+
+// Solve the state-dependent chain at arrival rate lam; all threads return
 // identical results.
 template <int NT>
-__device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N, int K,
+__device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int N, int K,
                                double *scratch) {
   const int tid = threadIdx.x;
   const double loglam = log(lam);
-  double tmax = -INFINITY;
-  for (int n = tid; n <= N; n += NT) tmax = fmax(tmax, (double)n * loglam - S[n]);
+  const int n0 = tid * g.chunk + 1;
+  const int n1 = min(n0 + g.chunk - 1, N);
+
+  // pass 1: max over anchor states (plus the n=0 state's t=0)
+  double tmax = 0.0;
+  for (int k = 0; k < g.ksub; ++k) {
+    int na = n0 + k * WVA_SUB;
+    if (na > n1) break;
+    tmax = fmax(tmax, (double)na * loglam - g.S_anchor[tid * g.ksub + k]);
+  }
   const double m = red_max<NT>(tmax, scratch);
-  double head_sum = 0.0, head_n_sum = 0.0;
-  for (int n = tid; n <= N; n += NT) {
-    double w = exp((double)n * loglam - S[n] - m);
+
+  // pass 2: head sums via per-sub-chunk running products
+  double head_sum = (tid == 0) ? exp(-m) : 0.0;  // n = 0 state
+  double head_n_sum = 0.0;
+  for (int k = 0; k < g.ksub; ++k) {
+    const int j0 = k * WVA_SUB;
+    const int na = n0 + j0;
+    if (na > n1) break;
+    double w = exp((double)na * loglam - g.S_anchor[tid * g.ksub + k] - m);
     head_sum += w;
-    head_n_sum += (double)n * w;
+    head_n_sum += (double)na * w;
+    const int jend = min(j0 + WVA_SUB - 1, n1 - n0);
+    for (int j = j0 + 1; j <= jend; ++j) {
+      w *= lam * g.inv_s_t[j * NT + tid];
+      head_sum += w;
+      head_n_sum += (double)(n0 + j) * w;
+    }
   }
   red_sum2<NT>(head_sum, head_n_sum, scratch);
 
   // geometric tail n = N+1..K with ratio r = lam/s(N)
   const double log_r = loglam - logsN;
   const double r = exp(log_r);
-  const double wN = exp((double)N * loglam - S[N] - m);
+  const double wN = exp((double)N * loglam - g.S_total - m);
   const int Q = K - N;
   double tail_sum = 0.0, tail_n_sum = 0.0, wK = (Q == 0) ? wN : 0.0;
   if (Q > 0 && wN > 0.0) {
@@ -220,10 +259,10 @@ __device__ ChainOut chain_eval(double lam, const double *S, double logsN, int N,
 }
 
 template <int NT>
-__device__ double eval_metric(int kind, double lam, const double *S, double logsN, int N, int K,
+__device__ double eval_metric(int kind, double lam, const ChainGeom &g, double logsN, int N, int K,
                               float gamma, float delta, float alpha, float beta, int in_tok,
                               int out_tok, double *scratch) {
-  ChainOut c = chain_eval<NT>(lam, S, logsN, N, K, scratch);
+  ChainOut c = chain_eval<NT>(lam, g, logsN, N, K, scratch);
   double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   if (kind == 0)
     return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
@@ -238,13 +277,13 @@ __device__ __forceinline__ bool within_tol(double x, double value) {
 
 // binary search matching pkg/analyzer/utils.go:26-70 (block in lockstep)
 template <int NT>
-__device__ double bisect(int kind, double x_min, double x_max, double y_target, const double *S,
+__device__ double bisect(int kind, double x_min, double x_max, double y_target, const ChainGeom &g,
                          double logsN, int N, int K, float gamma, float delta, float alpha,
                          float beta, int in_tok, int out_tok, double *scratch, int *ind) {
-  double y_lo = eval_metric<NT>(kind, x_min, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+  double y_lo = eval_metric<NT>(kind, x_min, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
                                 out_tok, scratch);
   if (within_tol(y_lo, y_target)) { *ind = 0; return x_min; }
-  double y_hi = eval_metric<NT>(kind, x_max, S, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+  double y_hi = eval_metric<NT>(kind, x_max, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
                                 out_tok, scratch);
   if (within_tol(y_hi, y_target)) { *ind = 0; return x_max; }
   const bool increasing = y_lo < y_hi;
@@ -259,7 +298,7 @@ __device__ double bisect(int kind, double x_min, double x_max, double y_target, 
   double x_star = 0.5 * (x_min + x_max);
   for (int it = 0; it < WVA_MAX_ITERS; ++it) {
     x_star = 0.5 * (x_min + x_max);
-    double y_star = eval_metric<NT>(kind, x_star, S, logsN, N, K, gamma, delta, alpha, beta,
+    double y_star = eval_metric<NT>(kind, x_star, g, logsN, N, K, gamma, delta, alpha, beta,
                                     in_tok, out_tok, scratch);
     if (within_tol(y_star, y_target)) break;
     if ((increasing && y_target < y_star) || (!increasing && y_target > y_star))
@@ -283,8 +322,10 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   if ((int)blockIdx.x >= n_blocks) return;
   const int cell = cell_ids ? cell_ids[blockIdx.x] : (int)blockIdx.x;
   const int tid = threadIdx.x;
-  double *S = smem;
-  double *scratch = smem + (max_n + 1);
+  // LDS layout: [0..31] reduction scratch, [32] S_total slot, [40..] chain
+  // geometry (inv_s_t then S_anchor); 40-double header keeps alignment.
+  double *scratch = smem;
+  double *S = smem + 40;  // geometry base (see setup below)
 
   const int in_tok = in.in_tok[cell];
   const int out_tok = in.out_tok[cell];
@@ -360,8 +401,14 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   int num_decode = out_tok - 1;
   if (in_tok == 0 && out_tok == 1) num_decode = 1;
 
-  // ---- build LDS log-prefix of service rates (chunked scan) ----
+  // ---- build chain geometry: transposed 1/s + anchor log-prefix ----
   const int chunk = (N + NT - 1) / NT;
+  const int ksub = (chunk + WVA_SUB - 1) / WVA_SUB;
+  // LDS partition (S points at the dynamic smem base; scratch follows)
+  double *inv_s_t = S;                        // chunk*NT doubles
+  double *S_anchor = inv_s_t + chunk * NT;    // NT*ksub doubles
+  double *total_slot = scratch + 32;
+
   const int n0 = tid * chunk + 1;
   const int n1 = min(n0 + chunk - 1, N);
   double local = 0.0;
@@ -371,7 +418,9 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
     float decode = alpha + beta * nf;
     float s = nf / (prefill + (float)num_decode * decode);  // fp32 like the reference
     local += log((double)s);
-    S[n] = local;  // chunk-local prefix for now
+    const int j = n - n0;
+    inv_s_t[j * NT + tid] = 1.0 / (double)s;
+    if ((j % WVA_SUB) == 0) S_anchor[tid * ksub + j / WVA_SUB] = local;  // chunk-local
   }
   // scan of per-thread chunk totals: intra-wave in registers...
   double incl = local;
@@ -391,13 +440,20 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
     offset += carry;
     static_assert(NW >= 1, "block must be at least one wave");
   }
-  for (int n = n0; n <= n1; ++n) S[n] += offset;
-  if (tid == 0) S[0] = 0.0;
+  for (int k = 0; k < ksub && n0 + k * WVA_SUB <= n1; ++k)
+    S_anchor[tid * ksub + k] += offset;
+  if (tid == NT - 1) *total_slot = offset + local;  // S[N]
   if constexpr (NT == WVA_WAVE) {
     __builtin_amdgcn_s_waitcnt(0);  // single wave: order LDS writes before reads
   } else {
     __syncthreads();
   }
+  ChainGeom geom;
+  geom.inv_s_t = inv_s_t;
+  geom.S_anchor = S_anchor;
+  geom.S_total = *total_slot;
+  geom.chunk = chunk;
+  geom.ksub = ksub;
 
   // s(1), s(N) in fp32 (rate-range bounds, ref queueanalyzer.go:117-119)
   float prefill1 = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok);
@@ -415,13 +471,13 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   bool feasible = true;
   double lam_ttft = lam_max;
   if (t_ttft > 0.0f) {
-    lam_ttft = bisect<NT>(0, lam_min, lam_max, (double)t_ttft, S, logsN, N, Kstates, gamma,
+    lam_ttft = bisect<NT>(0, lam_min, lam_max, (double)t_ttft, geom, logsN, N, Kstates, gamma,
                           delta, alpha, beta, in_tok, out_tok, scratch, &ind);
     if (ind < 0) feasible = false;
   }
   double lam_itl = lam_max;
   if (feasible && t_itl > 0.0f) {
-    lam_itl = bisect<NT>(1, lam_min, lam_max, (double)t_itl, S, logsN, N, Kstates, gamma, delta,
+    lam_itl = bisect<NT>(1, lam_min, lam_max, (double)t_itl, geom, logsN, N, Kstates, gamma, delta,
                          alpha, beta, in_tok, out_tok, scratch, &ind);
     if (ind < 0) feasible = false;
   }
@@ -436,7 +492,7 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
 
   // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
-  ChainOut c = chain_eval<NT>(lam, S, logsN, N, Kstates, scratch);
+  ChainOut c = chain_eval<NT>(lam, geom, logsN, N, Kstates, scratch);
   const double rate_star = c.throughput * 1000.0;  // req/sec
 
   double total_rate;  // req/sec (ref allocation.go:134-139)
@@ -454,7 +510,7 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
 
   // ---- per-replica analyze (ref allocation.go:148-157) ----
   const double rate = total_rate / (double)num_replicas;
-  ChainOut c2 = chain_eval<NT>(rate / 1000.0, S, logsN, N, Kstates, scratch);
+  ChainOut c2 = chain_eval<NT>(rate / 1000.0, geom, logsN, N, Kstates, scratch);
   const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
   const float token_t = decode_time_f(alpha, beta, (float)eff);
@@ -533,7 +589,9 @@ extern "C" int wva_sweep_launch_bucket(
                    cur_cost};
   WvaCellsOut out = {feasible, zero_empty, num_replicas, batch, cost, value, itl, ttft, rho,
                      max_rate};
-  size_t lds = (size_t)(max_n + 1 + 2 * (1024 / WVA_WAVE)) * sizeof(double);
+  const int chunk = (max_n + nt - 1) / nt;
+  const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
+  size_t lds = (size_t)(40 + chunk * nt + (size_t)nt * ksub) * sizeof(double);
   switch (nt) {
     case 64:
       hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,
