@@ -117,24 +117,65 @@ def main():
         load_library(allow_build=False)  # fail loudly if the in-tree .so is missing
     solver = ShardedSolver(engine)
 
+    import numpy as np
+
     all_names = sorted(system.servers)
     local_names = shard_servers(all_names, rank, world)
+    local_gidx = np.arange(rank, len(all_names), world)
     trace = PoissonTrace(len(all_names), seed=args.seed + 7)
-    in_toks = [system.servers[n].load.avgInTokens for n in all_names]
-    out_toks = [system.servers[n].load.avgOutTokens for n in all_names]
+    in_toks = np.array(
+        [system.servers[n].load.avgInTokens for n in all_names], dtype=np.int32
+    )
+    out_toks = np.array(
+        [system.servers[n].load.avgOutTokens for n in all_names], dtype=np.int32
+    )
+    # steady-state current-allocation arrays (penalty inputs), local shard
+    acc_index = {n: i for i, n in enumerate(sorted(system.accelerators))}
+    cur_acc = np.empty(len(local_names), dtype=np.int32)
+    cur_rep = np.zeros(len(local_names), dtype=np.int32)
+    cur_cost = np.zeros(len(local_names), dtype=np.float32)
+    for j, name in enumerate(local_names):
+        cur = system.servers[name].cur_allocation
+        cur_acc[j] = -3 if cur is None else acc_index.get(cur.accelerator, -1)
+        if cur is not None:
+            cur_rep[j] = cur.num_replicas
+            cur_cost[j] = cur.cost
+    state = {"cur": (cur_acc, cur_rep, cur_cost)}
 
     def reconcile(step: int):
         rates = trace.rates_at(step)
-        for i, name in enumerate(all_names):
-            system.servers[name].load = ServerLoadSpec(
-                arrivalRate=float(rates[i]), avgInTokens=in_toks[i], avgOutTokens=out_toks[i]
+        fs = solver.fast_sweep
+        if use_gpu and fs is not None:
+            # steady state: loads + currents flow as arrays (no object churn)
+            fs.load_override = (
+                rates[local_gidx].astype(np.float32),
+                in_toks[local_gidx],
+                out_toks[local_gidx],
             )
+            fs.cur_override = state["cur"]
+        else:
+            for i, name in enumerate(all_names):
+                system.servers[name].load = ServerLoadSpec(
+                    arrivalRate=float(rates[i]),
+                    avgInTokens=int(in_toks[i]),
+                    avgOutTokens=int(out_toks[i]),
+                )
         result = solver.solve(system, opt_spec)
         # apply: desired -> current (HPA/actuation convergence between ticks)
-        for name, data in result.solution.items():
-            server = system.servers[name]
-            server.spec.currentAlloc = data
-            server.cur_allocation = allocation_from_data(data)
+        w = result.winners
+        acc = w.acc_idx[local_gidx]
+        ok = w.valid[local_gidx] & (acc != -1)
+        pa, pr, pc = state["cur"]
+        state["cur"] = (
+            np.where(ok, np.where(acc == -2, -1, acc), pa).astype(np.int32),
+            np.where(ok, w.num_replicas[local_gidx], pr).astype(np.int32),
+            np.where(ok, w.cost[local_gidx], pc).astype(np.float32),
+        )
+        if not (use_gpu and fs is not None):
+            for name, data in result.solution.items():
+                server = system.servers[name]
+                server.spec.currentAlloc = data
+                server.cur_allocation = allocation_from_data(data)
         return result
 
     cells_per_step = None

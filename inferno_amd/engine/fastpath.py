@@ -112,27 +112,49 @@ class FastSweep:
         self._torch_cache = None
 
     # ------------------------------------------------------------------
+    # Array overrides for the steady-state loop: when set, the per-server
+    # Python object reads are skipped entirely (the bench/controller can
+    # maintain these arrays from winner records between reconciles).
+    # load_override = (arrival f32, in_tok i32, out_tok i32) per local server;
+    # cur_override = (cur_acc i32 [-3 none, -1 empty, >=0 acc index],
+    #                 cur_replicas i32, cur_cost f32).
+    load_override = None
+    cur_override = None
+
     def _refresh_dynamic(self) -> dict:
         """Gather per-server dynamic state into per-cell arrays (vectorized)."""
         n_srv = len(self.server_names)
-        arrival = np.empty(n_srv, dtype=np.float32)
-        in_tok = np.empty(n_srv, dtype=np.int32)
-        out_tok = np.empty(n_srv, dtype=np.int32)
-        cur_acc = np.empty(n_srv, dtype=np.int32)  # -1 empty, -3 no cur
-        cur_rep = np.zeros(n_srv, dtype=np.int32)
-        cur_cost = np.zeros(n_srv, dtype=np.float32)
-        for i, server in enumerate(self._srv_objs):
-            load = server.load
-            arrival[i] = load.arrivalRate if load is not None else 0.0
-            in_tok[i] = load.avgInTokens if load is not None else 0
-            out_tok[i] = load.avgOutTokens if load is not None else 0
-            cur = server.cur_allocation
-            if cur is None:
-                cur_acc[i] = -3
-            else:
-                cur_acc[i] = self._acc_index.get(cur.accelerator, -1)
-                cur_rep[i] = cur.num_replicas
-                cur_cost[i] = cur.cost
+        if self.load_override is not None:
+            arrival, in_tok, out_tok = self.load_override
+            arrival = np.asarray(arrival, dtype=np.float32)
+            in_tok = np.asarray(in_tok, dtype=np.int32)
+            out_tok = np.asarray(out_tok, dtype=np.int32)
+        else:
+            arrival = np.empty(n_srv, dtype=np.float32)
+            in_tok = np.empty(n_srv, dtype=np.int32)
+            out_tok = np.empty(n_srv, dtype=np.int32)
+            for i, server in enumerate(self._srv_objs):
+                load = server.load
+                arrival[i] = load.arrivalRate if load is not None else 0.0
+                in_tok[i] = load.avgInTokens if load is not None else 0
+                out_tok[i] = load.avgOutTokens if load is not None else 0
+        if self.cur_override is not None:
+            cur_acc, cur_rep, cur_cost = self.cur_override
+            cur_acc = np.asarray(cur_acc, dtype=np.int32)
+            cur_rep = np.asarray(cur_rep, dtype=np.int32)
+            cur_cost = np.asarray(cur_cost, dtype=np.float32)
+        else:
+            cur_acc = np.empty(n_srv, dtype=np.int32)  # -1 empty, -3 no cur
+            cur_rep = np.zeros(n_srv, dtype=np.int32)
+            cur_cost = np.zeros(n_srv, dtype=np.float32)
+            for i, server in enumerate(self._srv_objs):
+                cur = server.cur_allocation
+                if cur is None:
+                    cur_acc[i] = -3
+                else:
+                    cur_acc[i] = self._acc_index.get(cur.accelerator, -1)
+                    cur_rep[i] = cur.num_replicas
+                    cur_cost[i] = cur.cost
 
         cs = self.cell_server
         c_out = out_tok[cs]
@@ -246,23 +268,17 @@ class FastSweep:
     def _buckets_for(self, batch_n: np.ndarray):
         import torch
 
-        from ..ops.sweep import N_MED, N_SMALL
+        from ..ops.sweep import choose_buckets
 
         st = self._gpu
         key = batch_n.tobytes()
         if st["bucket_key"] == key:
             return st["buckets"]
-        buckets = []
-        small = batch_n <= N_SMALL
-        med = (batch_n > N_SMALL) & (batch_n <= N_MED)
-        large = batch_n > N_MED
-        for nt, mask in ((64, small), (256, med), (1024, large)):
-            idx = np.nonzero(mask)[0]
-            if len(idx):
-                ids = None
-                if len(idx) != len(batch_n):
-                    ids = torch.from_numpy(idx.astype(np.int32)).to(self.device)
-                buckets.append((nt, ids, int(batch_n[mask].max()), len(idx)))
+        buckets = [
+            (nt, torch.from_numpy(ids).to(self.device) if ids is not None else None,
+             bmax, count)
+            for nt, ids, bmax, count in choose_buckets(batch_n)
+        ]
         st["bucket_key"] = key
         st["buckets"] = buckets
         while len(st["side_streams"]) < max(len(buckets) - 1, 0):

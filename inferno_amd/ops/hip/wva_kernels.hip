@@ -178,7 +178,13 @@ struct ChainOut {
 #define WVA_SUB 32
 
 struct ChainGeom {
-  const double *inv_s_t;   // [chunk*NT] transposed reciprocal service rates
+  const float *inv_s_t;    // [chunk*NT] transposed reciprocal service rates
+                           // (fp32: the rates themselves are fp32 per the
+                           // reference; one extra rounding per factor is
+                           // reset at every 32-state anchor, keeping the
+                           // sub-chunk product within ~1e-6 relative — and
+                           // it halves the LDS footprint, doubling the
+                           // blocks/CU for large-N cells)
   const double *S_anchor;  // [NT*ksub] log-prefix at anchors
   double S_total;          // S[N]
   int chunk;               // states per lane = ceil(N/NT)
@@ -216,7 +222,7 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
     head_n_sum += (double)na * w;
     const int jend = min(j0 + WVA_SUB - 1, n1 - n0);
     for (int j = j0 + 1; j <= jend; ++j) {
-      w *= lam * g.inv_s_t[j * NT + tid];
+      w *= lam * (double)g.inv_s_t[j * NT + tid];
       head_sum += w;
       head_n_sum += (double)(n0 + j) * w;
     }
@@ -404,9 +410,9 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   // ---- build chain geometry: transposed 1/s + anchor log-prefix ----
   const int chunk = (N + NT - 1) / NT;
   const int ksub = (chunk + WVA_SUB - 1) / WVA_SUB;
-  // LDS partition (S points at the dynamic smem base; scratch follows)
-  double *inv_s_t = S;                        // chunk*NT doubles
-  double *S_anchor = inv_s_t + chunk * NT;    // NT*ksub doubles
+  // LDS partition (S points at the dynamic smem base; scratch precedes)
+  float *inv_s_t = (float *)S;                // chunk*NT floats
+  double *S_anchor = S + (chunk * NT + 1) / 2;  // NT*ksub doubles (8B aligned)
   double *total_slot = scratch + 32;
 
   const int n0 = tid * chunk + 1;
@@ -419,7 +425,7 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
     float s = nf / (prefill + (float)num_decode * decode);  // fp32 like the reference
     local += log((double)s);
     const int j = n - n0;
-    inv_s_t[j * NT + tid] = 1.0 / (double)s;
+    inv_s_t[j * NT + tid] = (float)(1.0 / (double)s);
     if ((j % WVA_SUB) == 0) S_anchor[tid * ksub + j / WVA_SUB] = local;  // chunk-local
   }
   // scan of per-thread chunk totals: intra-wave in registers...
@@ -591,7 +597,9 @@ extern "C" int wva_sweep_launch_bucket(
                      max_rate};
   const int chunk = (max_n + nt - 1) / nt;
   const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
-  size_t lds = (size_t)(40 + chunk * nt + (size_t)nt * ksub) * sizeof(double);
+  // header(40) + inv_s floats (chunk*nt/2 doubles, rounded up) + anchors
+  size_t lds =
+      (size_t)(40 + (chunk * nt + 1) / 2 + (size_t)nt * ksub) * sizeof(double);
   switch (nt) {
     case 64:
       hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,

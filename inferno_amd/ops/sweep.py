@@ -22,6 +22,43 @@ MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
 N_SMALL = 512
 N_MED = 2048
 
+
+def choose_buckets(batch_n):
+    """Partition cells into N-buckets and pick each bucket's block size.
+
+    Wide blocks shorten a single cell's latency (more lanes per chain pass)
+    but cost residency (a 1024-thread block is 1-4 blocks/CU); they pay off
+    only in the LATENCY regime, when the bucket has fewer cells than the
+    chip has resident slots. In the throughput regime (cells >> CUs) smaller
+    blocks win: more cells in flight at the same total work.
+
+    Returns [(nt, cell_idx int32 array or None, bucket_max_n, count)].
+    """
+    import numpy as np
+
+    n = len(batch_n)
+    masks = [
+        (64, batch_n <= N_SMALL, None),
+        (256, (batch_n > N_SMALL) & (batch_n <= N_MED), 64),
+        (1024, batch_n > N_MED, 256),
+    ]
+    out = []
+    for base_nt, mask, demoted in masks:
+        idx = np.nonzero(mask)[0]
+        count = len(idx)
+        if count == 0:
+            continue
+        nt = base_nt
+        # demote to the narrower block size when the bucket is
+        # throughput-bound (more blocks resident than the wide shape allows)
+        if demoted is not None:
+            resident = {256: 1024, 1024: 512}[base_nt]
+            if count > resident:
+                nt = demoted
+        ids = None if count == n else idx.astype(np.int32)
+        out.append((nt, ids, int(batch_n[mask].max()), count))
+    return out
+
 _lib: Optional[ctypes.CDLL] = None
 
 
@@ -103,23 +140,11 @@ def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
     )
     stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
 
-    # partition cells into N-buckets so stragglers get wide blocks
-    batch_cpu = arrays["batch_n"]
-    buckets = []  # (nt, ids tensor or None, bucket_max_n)
-    small = batch_cpu <= N_SMALL
-    med = (batch_cpu > N_SMALL) & (batch_cpu <= N_MED)
-    large = batch_cpu > N_MED
-    if bool(small.all()):
-        buckets.append((64, None, int(max_n)))
-    elif bool(med.all()):
-        buckets.append((256, None, int(max_n)))
-    elif bool(large.all()):
-        buckets.append((1024, None, int(max_n)))
-    else:
-        for nt, mask in ((64, small), (256, med), (1024, large)):
-            ids = mask.nonzero(as_tuple=False).flatten().to(t.int32)
-            if ids.numel():
-                buckets.append((nt, ids.to(device), int(batch_cpu[mask].max().item())))
+    # partition cells into regime-adaptive N-buckets
+    buckets = [
+        (nt, torch.from_numpy(ids).to(device) if ids is not None else None, bmax)
+        for nt, ids, bmax, _count in choose_buckets(arrays["batch_n"].numpy())
+    ]
 
     # overlap the bucket launches on separate HIP streams: wall time becomes
     # the straggler bucket's latency instead of the sum of all three

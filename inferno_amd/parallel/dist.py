@@ -13,30 +13,29 @@ with identical results to a single-process solve.
 
 Hot path: FastSweep (engine/fastpath.py) — static cell structure cached per
 fleet topology, vectorized dynamic refresh, winners-only materialization.
-Greedy (limited) mode falls back to the full SweepEngine candidate lists.
+The global solution is carried as numpy arrays (``GlobalWinners``); the
+per-server ``AllocationData`` dict is materialized lazily only when a
+consumer (the controller's status writer) asks for it. Greedy (limited)
+mode falls back to the full SweepEngine candidate lists.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass
+from dataclasses import dataclass, field
+from functools import cached_property
 from typing import Optional
+
+import numpy as np
 
 from ..config import AllocationData, OptimizerSpec, SaturationPolicy
 from ..core.system import AllocationByType, System
 from ..engine import SweepEngine
-from ..engine.fastpath import FastSweep, WinnerRecord
+from ..engine.fastpath import FastSweep, WinnerRecord, _empty_winner
 
 # winner record encoding: [server_global_idx, acc_code, num_replicas, cost,
 #                          batch, itl, ttft, valid]
 _REC_W = 8
 _ACC_EMPTY = -2.0  # zero-load empty allocation (accelerator "")
 _ACC_NONE = -1.0  # no feasible allocation
-
-
-@dataclass
-class ShardResult:
-    solution: dict[str, AllocationData]
-    allocation_by_type: dict[str, AllocationByType]
-    local_stats: object
 
 
 def shard_servers(all_names: list[str], rank: int, world: int) -> list[str]:
@@ -49,6 +48,59 @@ class _ShardStats:
     n_servers: int = 0
 
 
+@dataclass
+class GlobalWinners:
+    """Global solution as flat arrays (index = position in sorted names)."""
+
+    names: list[str]
+    acc_names: list[str]
+    acc_idx: np.ndarray  # int32: accelerator index, -1 none, -2 empty
+    num_replicas: np.ndarray  # int32
+    batch: np.ndarray  # int32
+    cost: np.ndarray  # float32
+    itl: np.ndarray
+    ttft: np.ndarray
+    valid: np.ndarray  # bool: a rank reported this server
+
+
+@dataclass
+class ShardResult:
+    winners: GlobalWinners
+    allocation_by_type: dict[str, AllocationByType]
+    local_stats: object
+    _system: Optional[System] = field(default=None, repr=False)
+
+    @cached_property
+    def solution(self) -> dict[str, AllocationData]:
+        """Per-server AllocationData map (GenerateSolution wire shape);
+        materialized lazily from the winner arrays."""
+        w = self.winners
+        out: dict[str, AllocationData] = {}
+        for i, name in enumerate(w.names):
+            if not w.valid[i]:
+                continue
+            code = int(w.acc_idx[i])
+            if code == -1:
+                continue
+            acc = "" if code == -2 else w.acc_names[code]
+            load = None
+            if self._system is not None:
+                srv = self._system.servers.get(name)
+                load = srv.load if srv is not None else None
+            data = AllocationData(
+                accelerator=acc,
+                numReplicas=int(w.num_replicas[i]),
+                maxBatch=int(w.batch[i]),
+                cost=float(w.cost[i]),
+                itlAverage=float(w.itl[i]),
+                ttftAverage=float(w.ttft[i]),
+            )
+            if load is not None:
+                data.load = load
+            out[name] = data
+        return out
+
+
 class ShardedSolver:
     """Data-parallel sweep across ranks with all-gathered winners."""
 
@@ -58,11 +110,19 @@ class ShardedSolver:
         self.fast = fast
         self._fast_sweep: Optional[FastSweep] = None
         self._fast_key = None
+        self._agg_key = None
+        self._agg = None  # cached aggregation structures
 
     def invalidate(self) -> None:
-        """Drop the cached cell structure (fleet topology changed)."""
+        """Drop cached structures (fleet topology changed)."""
         self._fast_sweep = None
         self._fast_key = None
+        self._agg_key = None
+        self._agg = None
+
+    @property
+    def fast_sweep(self) -> Optional[FastSweep]:
+        return self._fast_sweep
 
     def _dist_info(self):
         import torch.distributed as dist
@@ -89,75 +149,107 @@ class ShardedSolver:
 
         # encode local winners into a fixed-size record tensor
         max_shard = (len(all_names) + world - 1) // world
-        rec_t = torch.full((max_shard, _REC_W), -3.0, dtype=torch.float32)
-        name_to_global = {n: i for i, n in enumerate(all_names)}
-        import numpy as np
-
         n_local = len(local_names)
+        body = np.full((max_shard, _REC_W), -3.0, dtype=np.float32)
         if n_local:
-            gidx = np.asarray([name_to_global[n] for n in local_names], dtype=np.float32)
-            body = np.stack(
-                [
-                    gidx,
-                    rec.acc_idx.astype(np.float32),
-                    rec.num_replicas.astype(np.float32),
-                    rec.cost,
-                    rec.batch.astype(np.float32),
-                    rec.itl,
-                    rec.ttft,
-                    np.ones(n_local, dtype=np.float32),
-                ],
-                axis=1,
-            )
-            rec_t[:n_local] = torch.from_numpy(body)
+            gidx = np.arange(rank, rank + world * n_local, world, dtype=np.float32)
+            body[:n_local, 0] = gidx[:n_local]
+            body[:n_local, 1] = rec.acc_idx
+            body[:n_local, 2] = rec.num_replicas
+            body[:n_local, 3] = rec.cost
+            body[:n_local, 4] = rec.batch
+            body[:n_local, 5] = rec.itl
+            body[:n_local, 6] = rec.ttft
+            body[:n_local, 7] = 1.0
 
         if initialized and world > 1:
             backend = dist.get_backend(self.group)
             comm_dev = "cuda" if backend == "nccl" else "cpu"
-            rec_d = rec_t.to(comm_dev)
+            rec_d = torch.from_numpy(body).to(comm_dev)
             gathered = [torch.empty_like(rec_d) for _ in range(world)]
             dist.all_gather(gathered, rec_d, group=self.group)
             all_rec = torch.cat(gathered, dim=0).cpu().numpy()
         else:
-            all_rec = rec_t.numpy()
+            all_rec = body
 
-        # reconstruct the global solution (identical on every rank)
-        solution: dict[str, AllocationData] = {}
-        by_type: dict[str, AllocationByType] = {}
-        for row in all_rec:
-            if row[7] != 1.0:
-                continue
-            name = all_names[int(row[0])]
-            code = row[1]
-            if code == _ACC_NONE:
-                continue
-            acc_name = "" if code == _ACC_EMPTY else acc_names[int(code)]
-            server = system.servers[name]
-            data = AllocationData(
-                accelerator=acc_name,
-                numReplicas=int(row[2]),
-                maxBatch=int(row[4]),
-                cost=float(row[3]),
-                itlAverage=float(row[5]),
-                ttftAverage=float(row[6]),
-                load=server.load,
-            )
-            solution[name] = data
-            if acc_name:
-                acc = system.accelerators.get(acc_name)
-                model = system.models.get(server.model_name)
-                if acc is not None and model is not None:
-                    t = acc.type
-                    agg = by_type.setdefault(
-                        t, AllocationByType(name=t, limit=system.capacity.get(t, 0))
-                    )
-                    agg.count += (
-                        data.numReplicas * model.get_num_instances(acc.name) * acc.multiplicity
-                    )
-                    agg.cost += data.cost
+        # scatter rows back into global order (vectorized)
+        n = len(all_names)
+        winners = GlobalWinners(
+            names=all_names,
+            acc_names=acc_names,
+            acc_idx=np.full(n, -1, dtype=np.int32),
+            num_replicas=np.zeros(n, dtype=np.int32),
+            batch=np.zeros(n, dtype=np.int32),
+            cost=np.zeros(n, dtype=np.float32),
+            itl=np.zeros(n, dtype=np.float32),
+            ttft=np.zeros(n, dtype=np.float32),
+            valid=np.zeros(n, dtype=bool),
+        )
+        ok = all_rec[:, 7] == 1.0
+        rows = all_rec[ok]
+        g = rows[:, 0].astype(np.int64)
+        winners.acc_idx[g] = rows[:, 1].astype(np.int32)
+        winners.num_replicas[g] = rows[:, 2].astype(np.int32)
+        winners.batch[g] = rows[:, 4].astype(np.int32)
+        winners.cost[g] = rows[:, 3]
+        winners.itl[g] = rows[:, 5]
+        winners.ttft[g] = rows[:, 6]
+        winners.valid[g] = True
 
+        by_type = self._aggregate_by_type(system, all_names, acc_names, winners)
         system.allocation_by_type = by_type
-        return ShardResult(solution=solution, allocation_by_type=by_type, local_stats=stats)
+        return ShardResult(
+            winners=winners,
+            allocation_by_type=by_type,
+            local_stats=stats,
+            _system=system,
+        )
+
+    # ------------------------------------------------------------------
+    def _aggregate_by_type(self, system, all_names, acc_names, winners):
+        """Vectorized AllocateByType (ref system.go:271-300)."""
+        key = (id(system), tuple(acc_names), len(all_names))
+        if self._agg_key != key:
+            n_acc = len(acc_names)
+            inst = np.zeros((len(all_names), n_acc), dtype=np.int32)
+            for i, name in enumerate(all_names):
+                model = system.models.get(system.servers[name].model_name)
+                if model is None:
+                    continue
+                for j, an in enumerate(acc_names):
+                    inst[i, j] = model.get_num_instances(an)
+            mult = np.array(
+                [system.accelerators[a].multiplicity for a in acc_names], dtype=np.int32
+            )
+            types = [system.accelerators[a].type for a in acc_names]
+            uniq_types = sorted(set(types))
+            type_idx = np.array([uniq_types.index(t) for t in types], dtype=np.int32)
+            self._agg = (inst, mult, types, uniq_types, type_idx)
+            self._agg_key = key
+        inst, mult, types, uniq_types, type_idx = self._agg
+
+        sel = winners.valid & (winners.acc_idx >= 0)
+        idx = np.nonzero(sel)[0]
+        by_type: dict[str, AllocationByType] = {}
+        if len(idx):
+            codes = winners.acc_idx[idx]
+            counts = (
+                winners.num_replicas[idx] * inst[idx, codes] * mult[codes]
+            ).astype(np.int64)
+            tcount = np.zeros(len(uniq_types), dtype=np.int64)
+            tcost = np.zeros(len(uniq_types), dtype=np.float64)
+            np.add.at(tcount, type_idx[codes], counts)
+            np.add.at(tcost, type_idx[codes], winners.cost[idx].astype(np.float64))
+            for k, t in enumerate(uniq_types):
+                if tcount[k] == 0 and tcost[k] == 0.0:
+                    continue
+                by_type[t] = AllocationByType(
+                    name=t,
+                    count=int(tcount[k]),
+                    limit=system.capacity.get(t, 0),
+                    cost=float(tcost[k]),
+                )
+        return by_type
 
     # ------------------------------------------------------------------
     def _solve_fast(self, system: System, local_names: list[str]):
@@ -201,10 +293,7 @@ class ShardedSolver:
             )
 
         acc_index = {n: i for i, n in enumerate(acc_names)}
-        n_local = len(local_names)
-        from ..engine.fastpath import _empty_winner
-
-        rec = _empty_winner(n_local)
+        rec = _empty_winner(len(local_names))
         for j, name in enumerate(local_names):
             alloc = system.servers[name].allocation
             if alloc is None:
