@@ -12,7 +12,7 @@ import json
 import math
 import os
 from dataclasses import dataclass
-from typing import Any, Optional
+from typing import Optional
 
 import yaml
 

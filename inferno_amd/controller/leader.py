@@ -5,7 +5,6 @@ from __future__ import annotations
 
 import time
 from datetime import datetime, timezone
-from typing import Optional
 
 
 def _now() -> str:

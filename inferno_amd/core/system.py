@@ -7,7 +7,7 @@ singleton (``core.TheSystem``, system.go:10-45) — every operation takes the
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 from ..config import (

@@ -14,7 +14,7 @@ on a virtual clock; the FastAPI server wraps it with a real-time loop.
 from __future__ import annotations
 
 import itertools
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Optional
 
 # defaults mirror tools/vllm-emulator/server.py:22-33

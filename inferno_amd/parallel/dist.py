@@ -29,7 +29,7 @@ import numpy as np
 from ..config import AllocationData, OptimizerSpec, SaturationPolicy
 from ..core.system import AllocationByType, System
 from ..engine import SweepEngine
-from ..engine.fastpath import FastSweep, WinnerRecord, _empty_winner
+from ..engine.fastpath import FastSweep, _empty_winner
 
 # winner record encoding: [server_global_idx, acc_code, num_replicas, cost,
 #                          batch, itl, ttft, valid]

@@ -12,7 +12,6 @@ sorted candidate allocations — comes from the GPU sweep; the tail is small.
 from __future__ import annotations
 
 import bisect
-import math
 from dataclasses import dataclass, field
 
 from ..config import SaturationPolicy
