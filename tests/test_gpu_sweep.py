@@ -224,3 +224,27 @@ class TestGreedyOnGpuSweep:
             if (a.servers[n].allocation is None) == (b.servers[n].allocation is None)
         )
         assert n_same >= len(a.servers) - 1
+
+
+class TestControllerOnGpu:
+    def test_reconcile_with_gpu_backend(self):
+        """Full controller reconcile with the HIP sweep as the solver."""
+        from prometheus_client import CollectorRegistry
+
+        from inferno_amd.api import v1alpha1 as api
+        from inferno_amd.controller.metrics import MetricsEmitter
+        from inferno_amd.controller.reconciler import Reconciler
+        from tests.test_controller import build_world
+
+        kube, prom, em, reg, _ = build_world(arrival_per_sec=6.0)
+        rec = Reconciler(
+            kube, prom, MetricsEmitter(registry=CollectorRegistry()),
+            backend="gpu", scale_to_zero=False,
+        )
+        result = rec.reconcile()
+        assert result.processed == 1
+        assert result.solver_backend == "gpu"
+        assert result.degraded is False
+        va = kube.vas[("default", "vllme-deploy")]
+        assert va.status.desiredOptimizedAlloc.numReplicas >= 1
+        assert api.is_condition_true(va, api.TYPE_OPTIMIZATION_READY)
