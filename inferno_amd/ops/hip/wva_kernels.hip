@@ -84,51 +84,77 @@ struct WvaCellsOut {
 };
 
 // ---------------------------------------------------------------------------
-// reductions templated on block size; scratch is LDS (>= 2*NT/WVA_WAVE doubles)
+// reductions templated on block size and PART (1 = whole block, 2 = the two
+// half-blocks reduce independently — used by the concurrent dual bisection).
+// scratch is LDS (>= 32 doubles); parts use disjoint scratch regions, and the
+// __syncthreads calls are executed uniformly by the whole block (both parts
+// run their evaluations in lock-step).
 // ---------------------------------------------------------------------------
-template <int NT>
-__device__ __forceinline__ double red_max(double v, double *scratch) {
+template <int NT, int PART>
+__device__ __forceinline__ double red_max_p(double v, double *scratch) {
+  constexpr int W = NT / PART;  // threads per part
+  if constexpr (W < WVA_WAVE) {
+    // sub-wave part (NT=64, PART=2): 32-lane butterfly, barrier-free
 #pragma unroll
-  for (int off = WVA_WAVE / 2; off > 0; off >>= 1) v = fmax(v, __shfl_xor(v, off, WVA_WAVE));
-  if constexpr (NT == WVA_WAVE) {
+    for (int off = W / 2; off > 0; off >>= 1) v = fmax(v, __shfl_xor(v, off, W));
     return v;
   } else {
-    constexpr int NW = NT / WVA_WAVE;
-    const int tid = threadIdx.x;
-    __syncthreads();  // protect scratch from previous use
-    if ((tid & (WVA_WAVE - 1)) == 0) scratch[tid / WVA_WAVE] = v;
-    __syncthreads();
-    double m = scratch[0];
 #pragma unroll
-    for (int i = 1; i < NW; ++i) m = fmax(m, scratch[i]);
-    return m;
+    for (int off = WVA_WAVE / 2; off > 0; off >>= 1)
+      v = fmax(v, __shfl_xor(v, off, WVA_WAVE));
+    constexpr int NWH = W / WVA_WAVE;  // waves per part
+    if constexpr (NWH == 1) {
+      return v;
+    } else {
+      const int tid = threadIdx.x;
+      const int part = tid / W;
+      const int widx = (tid % W) / WVA_WAVE;
+      __syncthreads();  // protect scratch from previous use
+      if ((tid & (WVA_WAVE - 1)) == 0) scratch[part * NWH + widx] = v;
+      __syncthreads();
+      double m = scratch[part * NWH];
+#pragma unroll
+      for (int i = 1; i < NWH; ++i) m = fmax(m, scratch[part * NWH + i]);
+      return m;
+    }
   }
 }
 
-template <int NT>
-__device__ __forceinline__ void red_sum2(double &a, double &b, double *scratch) {
+template <int NT, int PART>
+__device__ __forceinline__ void red_sum2_p(double &a, double &b, double *scratch) {
+  constexpr int W = NT / PART;
+  if constexpr (W < WVA_WAVE) {
 #pragma unroll
-  for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
-    a += __shfl_xor(a, off, WVA_WAVE);
-    b += __shfl_xor(b, off, WVA_WAVE);
-  }
-  if constexpr (NT != WVA_WAVE) {
-    constexpr int NW = NT / WVA_WAVE;
-    const int tid = threadIdx.x;
-    __syncthreads();
-    if ((tid & (WVA_WAVE - 1)) == 0) {
-      scratch[2 * (tid / WVA_WAVE)] = a;
-      scratch[2 * (tid / WVA_WAVE) + 1] = b;
+    for (int off = W / 2; off > 0; off >>= 1) {
+      a += __shfl_xor(a, off, W);
+      b += __shfl_xor(b, off, W);
     }
-    __syncthreads();
-    double sa = 0.0, sb = 0.0;
+  } else {
 #pragma unroll
-    for (int i = 0; i < NW; ++i) {
-      sa += scratch[2 * i];
-      sb += scratch[2 * i + 1];
+    for (int off = WVA_WAVE / 2; off > 0; off >>= 1) {
+      a += __shfl_xor(a, off, WVA_WAVE);
+      b += __shfl_xor(b, off, WVA_WAVE);
     }
-    a = sa;
-    b = sb;
+    constexpr int NWH = W / WVA_WAVE;
+    if constexpr (NWH > 1) {
+      const int tid = threadIdx.x;
+      const int part = tid / W;
+      const int widx = (tid % W) / WVA_WAVE;
+      __syncthreads();
+      if ((tid & (WVA_WAVE - 1)) == 0) {
+        scratch[part * 2 * NWH + 2 * widx] = a;
+        scratch[part * 2 * NWH + 2 * widx + 1] = b;
+      }
+      __syncthreads();
+      double sa = 0.0, sb = 0.0;
+#pragma unroll
+      for (int i = 0; i < NWH; ++i) {
+        sa += scratch[part * 2 * NWH + 2 * i];
+        sb += scratch[part * 2 * NWH + 2 * i + 1];
+      }
+      a = sa;
+      b = sb;
+    }
   }
 }
 
@@ -191,43 +217,53 @@ struct ChainGeom {
   int ksub;                // anchors per lane = ceil(chunk/SUB)
 };
 
-// Solve the state-dependent chain at arrival rate lam; all threads return
-// identical results.
-template <int NT>
+// Solve the state-dependent chain at arrival rate lam. With PART=1 the whole
+// block cooperates and every thread returns identical results; with PART=2
+// each half-block independently evaluates its own lam (each half covers all
+// N states by walking two lane-chunks), enabling the concurrent TTFT/ITL
+// bisections.
+template <int NT, int PART>
 __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int N, int K,
                                double *scratch) {
-  const int tid = threadIdx.x;
+  constexpr int W = NT / PART;
+  const int lane = threadIdx.x % W;
   const double loglam = log(lam);
-  const int n0 = tid * g.chunk + 1;
-  const int n1 = min(n0 + g.chunk - 1, N);
 
   // pass 1: max over anchor states (plus the n=0 state's t=0)
   double tmax = 0.0;
-  for (int k = 0; k < g.ksub; ++k) {
-    int na = n0 + k * WVA_SUB;
-    if (na > n1) break;
-    tmax = fmax(tmax, (double)na * loglam - g.S_anchor[tid * g.ksub + k]);
-  }
-  const double m = red_max<NT>(tmax, scratch);
-
-  // pass 2: head sums via per-sub-chunk running products
-  double head_sum = (tid == 0) ? exp(-m) : 0.0;  // n = 0 state
-  double head_n_sum = 0.0;
-  for (int k = 0; k < g.ksub; ++k) {
-    const int j0 = k * WVA_SUB;
-    const int na = n0 + j0;
-    if (na > n1) break;
-    double w = exp((double)na * loglam - g.S_anchor[tid * g.ksub + k] - m);
-    head_sum += w;
-    head_n_sum += (double)na * w;
-    const int jend = min(j0 + WVA_SUB - 1, n1 - n0);
-    for (int j = j0 + 1; j <= jend; ++j) {
-      w *= lam * (double)g.inv_s_t[j * NT + tid];
-      head_sum += w;
-      head_n_sum += (double)(n0 + j) * w;
+  for (int c = lane; c < NT; c += W) {
+    const int n0c = c * g.chunk + 1;
+    const int n1c = min(n0c + g.chunk - 1, N);
+    for (int k = 0; k < g.ksub; ++k) {
+      int na = n0c + k * WVA_SUB;
+      if (na > n1c) break;
+      tmax = fmax(tmax, (double)na * loglam - g.S_anchor[c * g.ksub + k]);
     }
   }
-  red_sum2<NT>(head_sum, head_n_sum, scratch);
+  const double m = red_max_p<NT, PART>(tmax, scratch);
+
+  // pass 2: head sums via per-sub-chunk running products
+  double head_sum = (lane == 0) ? exp(-m) : 0.0;  // n = 0 state
+  double head_n_sum = 0.0;
+  for (int c = lane; c < NT; c += W) {
+    const int n0c = c * g.chunk + 1;
+    const int n1c = min(n0c + g.chunk - 1, N);
+    for (int k = 0; k < g.ksub; ++k) {
+      const int j0 = k * WVA_SUB;
+      const int na = n0c + j0;
+      if (na > n1c) break;
+      double w = exp((double)na * loglam - g.S_anchor[c * g.ksub + k] - m);
+      head_sum += w;
+      head_n_sum += (double)na * w;
+      const int jend = min(j0 + WVA_SUB - 1, n1c - n0c);
+      for (int j = j0 + 1; j <= jend; ++j) {
+        w *= lam * (double)g.inv_s_t[j * NT + c];
+        head_sum += w;
+        head_n_sum += (double)(n0c + j) * w;
+      }
+    }
+  }
+  red_sum2_p<NT, PART>(head_sum, head_n_sum, scratch);
 
   // geometric tail n = N+1..K with ratio r = lam/s(N)
   const double log_r = loglam - logsN;
@@ -243,10 +279,10 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
     } else {
       const double rQ = exp((double)Q * log_r);
       const double omr = 1.0 - r;
-      const double g = r * (1.0 - rQ) / omr;
+      const double gg = r * (1.0 - rQ) / omr;
       const double jg = r * (1.0 - (double)(Q + 1) * rQ + (double)Q * rQ * r) / (omr * omr);
-      tail_sum = wN * g;
-      tail_n_sum = wN * ((double)N * g + jg);
+      tail_sum = wN * gg;
+      tail_n_sum = wN * ((double)N * gg + jg);
       wK = wN * rQ;
     }
   }
@@ -264,11 +300,11 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
   return o;
 }
 
-template <int NT>
+template <int NT, int PART>
 __device__ double eval_metric(int kind, double lam, const ChainGeom &g, double logsN, int N, int K,
                               float gamma, float delta, float alpha, float beta, int in_tok,
                               int out_tok, double *scratch) {
-  ChainOut c = chain_eval<NT>(lam, g, logsN, N, K, scratch);
+  ChainOut c = chain_eval<NT, PART>(lam, g, logsN, N, K, scratch);
   double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   if (kind == 0)
     return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
@@ -281,39 +317,108 @@ __device__ __forceinline__ bool within_tol(double x, double value) {
   return fabs((x - value) / value) <= WVA_TOL;
 }
 
-// binary search matching pkg/analyzer/utils.go:26-70 (block in lockstep)
+// Concurrent dual bisection: half-block 0 searches the TTFT target, half 1
+// the ITL target, both matching pkg/analyzer/utils.go:26-70 exactly (boundary
+// tolerance checks, below/above indicators, <=100 midpoint iterations with
+// relative-tolerance early exit). The halves advance in LOCK-STEP — each
+// step both halves run exactly one chain evaluation (a finished half keeps
+// evaluating its last point as a no-op) so the __syncthreads counts inside
+// the partial reductions stay uniform; a block-wide vote ends the loop when
+// both halves are done. This halves the sequential depth of the sizing
+// phase vs running the two searches back to back.
 template <int NT>
-__device__ double bisect(int kind, double x_min, double x_max, double y_target, const ChainGeom &g,
-                         double logsN, int N, int K, float gamma, float delta, float alpha,
-                         float beta, int in_tok, int out_tok, double *scratch, int *ind) {
-  double y_lo = eval_metric<NT>(kind, x_min, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                                out_tok, scratch);
-  if (within_tol(y_lo, y_target)) { *ind = 0; return x_min; }
-  double y_hi = eval_metric<NT>(kind, x_max, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                                out_tok, scratch);
-  if (within_tol(y_hi, y_target)) { *ind = 0; return x_max; }
-  const bool increasing = y_lo < y_hi;
-  if ((increasing && y_target < y_lo) || (!increasing && y_target > y_lo)) {
-    *ind = -1;
-    return x_min;
-  }
-  if ((increasing && y_target > y_hi) || (!increasing && y_target < y_hi)) {
-    *ind = +1;
-    return x_max;
-  }
-  double x_star = 0.5 * (x_min + x_max);
-  for (int it = 0; it < WVA_MAX_ITERS; ++it) {
-    x_star = 0.5 * (x_min + x_max);
-    double y_star = eval_metric<NT>(kind, x_star, g, logsN, N, K, gamma, delta, alpha, beta,
-                                    in_tok, out_tok, scratch);
-    if (within_tol(y_star, y_target)) break;
-    if ((increasing && y_target < y_star) || (!increasing && y_target > y_star))
-      x_max = x_star;
+__device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float t_itl,
+                            const ChainGeom &g, double logsN, int N, int K, float gamma,
+                            float delta, float alpha, float beta, int in_tok, int out_tok,
+                            double *scratch, double *res_slot, double lam_star[2],
+                            int ind_out[2]) {
+  constexpr int W = NT / 2;
+  const int tid = threadIdx.x;
+  const int h = tid / W;  // 0 = TTFT, 1 = ITL
+  const double target = (h == 0) ? (double)t_ttft : (double)t_itl;
+  const bool active = target > 0.0;
+
+  double x_min = lam_min, x_max = lam_max;
+  double x_star = lam_max;
+  double y_lo = 0.0;
+  bool increasing = true;
+  int ind = 0;
+  int phase = active ? 0 : 3;  // 0 eval lo, 1 eval hi, 2 bisect, 3 done
+  int iters = 0;
+
+  for (int step = 0; step < WVA_MAX_ITERS + 3; ++step) {
+    double x_eval;
+    if (phase == 0)
+      x_eval = x_min;
+    else if (phase == 1)
+      x_eval = x_max;
+    else if (phase == 2)
+      x_eval = 0.5 * (x_min + x_max);
     else
-      x_min = x_star;
+      x_eval = x_max;  // done: dummy evaluation to keep barriers uniform
+    const double y =
+        eval_metric<NT, 2>(h, x_eval, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
+                           out_tok, scratch);
+    if (phase == 0) {
+      y_lo = y;
+      if (within_tol(y_lo, target)) {
+        x_star = x_min;
+        phase = 3;
+      } else {
+        phase = 1;
+      }
+    } else if (phase == 1) {
+      const double y_hi = y;
+      if (within_tol(y_hi, target)) {
+        x_star = x_max;
+        phase = 3;
+      } else {
+        increasing = y_lo < y_hi;
+        if ((increasing && target < y_lo) || (!increasing && target > y_lo)) {
+          ind = -1;
+          x_star = x_min;
+          phase = 3;
+        } else if ((increasing && target > y_hi) || (!increasing && target < y_hi)) {
+          ind = +1;
+          x_star = x_max;
+          phase = 3;
+        } else {
+          phase = 2;
+        }
+      }
+    } else if (phase == 2) {
+      x_star = x_eval;
+      ++iters;
+      if (within_tol(y, target) || iters >= WVA_MAX_ITERS) {
+        phase = 3;
+      } else if ((increasing && target < y) || (!increasing && target > y)) {
+        x_max = x_star;
+      } else {
+        x_min = x_star;
+      }
+    }
+    const bool done = (phase == 3);
+    if constexpr (NT == WVA_WAVE) {
+      if (__all(done ? 1 : 0)) break;
+    } else {
+      if (__syncthreads_and(done ? 1 : 0)) break;
+    }
   }
-  *ind = 0;
-  return x_star;
+
+  // publish both halves' results (res_slot: 4 doubles in LDS)
+  if ((tid % W) == 0) {
+    res_slot[2 * h] = x_star;
+    res_slot[2 * h + 1] = (double)ind;
+  }
+  if constexpr (NT == WVA_WAVE) {
+    __builtin_amdgcn_s_waitcnt(0);
+  } else {
+    __syncthreads();
+  }
+  lam_star[0] = res_slot[0];
+  ind_out[0] = (int)res_slot[1];
+  lam_star[1] = res_slot[2];
+  ind_out[1] = (int)res_slot[3];
 }
 
 // ---------------------------------------------------------------------------
@@ -472,20 +577,21 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   const double lam_max = (double)sN * (1.0 - WVA_EPSILON);
   const int Kstates = 11 * N;  // maxQueue (10N) + N  (ref allocation.go:87)
 
-  // ---- SLO sizing: TTFT and ITL bisections (ref queueanalyzer.go:185-255) --
-  int ind = 0;
+  // ---- SLO sizing: concurrent TTFT+ITL bisections (queueanalyzer.go:185-255)
+  double lam_star[2];
+  int inds[2];
+  dual_bisect<NT>(lam_min, lam_max, t_ttft, t_itl, geom, logsN, N, Kstates, gamma, delta,
+                  alpha, beta, in_tok, out_tok, scratch, total_slot + 1, lam_star, inds);
   bool feasible = true;
   double lam_ttft = lam_max;
   if (t_ttft > 0.0f) {
-    lam_ttft = bisect<NT>(0, lam_min, lam_max, (double)t_ttft, geom, logsN, N, Kstates, gamma,
-                          delta, alpha, beta, in_tok, out_tok, scratch, &ind);
-    if (ind < 0) feasible = false;
+    lam_ttft = lam_star[0];
+    if (inds[0] < 0) feasible = false;
   }
   double lam_itl = lam_max;
   if (feasible && t_itl > 0.0f) {
-    lam_itl = bisect<NT>(1, lam_min, lam_max, (double)t_itl, geom, logsN, N, Kstates, gamma, delta,
-                         alpha, beta, in_tok, out_tok, scratch, &ind);
-    if (ind < 0) feasible = false;
+    lam_itl = lam_star[1];
+    if (inds[1] < 0) feasible = false;
   }
   if (!feasible) {
     if (tid == 0) {
@@ -498,7 +604,7 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
   double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
 
   // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
-  ChainOut c = chain_eval<NT>(lam, geom, logsN, N, Kstates, scratch);
+  ChainOut c = chain_eval<NT, 1>(lam, geom, logsN, N, Kstates, scratch);
   const double rate_star = c.throughput * 1000.0;  // req/sec
 
   double total_rate;  // req/sec (ref allocation.go:134-139)
@@ -516,7 +622,7 @@ __global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out
 
   // ---- per-replica analyze (ref allocation.go:148-157) ----
   const double rate = total_rate / (double)num_replicas;
-  ChainOut c2 = chain_eval<NT>(rate / 1000.0, geom, logsN, N, Kstates, scratch);
+  ChainOut c2 = chain_eval<NT, 1>(rate / 1000.0, geom, logsN, N, Kstates, scratch);
   const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
   const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
   const float token_t = decode_time_f(alpha, beta, (float)eff);
