@@ -392,7 +392,7 @@ class FastSweep:
         pin_f[1] = torch.from_numpy(np.ascontiguousarray(arrs["cur_cost"], dtype=np.float32))
 
         buckets = self._buckets_for(arrs["batch_n"])
-        use_graph = os.environ.get("INFERNO_HIPGRAPH", "1") == "1"
+        use_graph = os.environ.get("INFERNO_HIPGRAPH", "0") == "1"  # measured: parity with eager (the launch path is already ~5 launches on pipelined streams); opt-in
         if use_graph and st.get("graph_key") == st["bucket_key"]:
             st["graph"].replay()
         else:
