@@ -248,6 +248,13 @@ def add_metrics_to_opt_status(
     )
 
 
+def collect_inventory_k8s(kube=None) -> dict[str, dict[str, dict]]:
+    """Stub for future limited-mode support — WVA operates in unlimited mode
+    (ref internal/collector/collector.go:37-42 CollectInventoryK8S). The GPU
+    vendor label prefixes it will scan: nvidia.com, amd.com, intel.com."""
+    return {}
+
+
 class MockPromAPI:
     """Test double mirroring the reference's MockPromAPI
     (test/utils/unitutils.go:137-159): query->samples map with a default

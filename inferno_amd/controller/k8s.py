@@ -184,6 +184,34 @@ class HttpKube:
         )
         r.raise_for_status()
 
+    def watch_events(self, resource_version: str = "", timeout_seconds: int = 60):
+        """Stream watch events for VariantAutoscalings (the reference's
+        controller-runtime For(VA) watch with a create-only event filter,
+        controller.go:456-487). Yields (event_type, VariantAutoscaling).
+        """
+        params = {"watch": "1", "timeoutSeconds": str(timeout_seconds)}
+        if resource_version:
+            params["resourceVersion"] = resource_version
+        import json as _json
+
+        with self._client.stream(
+            "GET", f"/apis/{api.GROUP}/{api.VERSION}/{api.PLURAL}", params=params,
+            timeout=timeout_seconds + 10,
+        ) as r:
+            r.raise_for_status()
+            for line in r.iter_lines():
+                if not line:
+                    continue
+                try:
+                    evt = _json.loads(line)
+                except ValueError:
+                    continue
+                etype = evt.get("type", "")
+                obj = evt.get("object", {}) or {}
+                if obj.get("kind") != api.KIND:
+                    continue
+                yield etype, api.va_from_json(obj)
+
     def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
         path = f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/{api.PLURAL}/{va.name}"
         ref = {

@@ -117,14 +117,31 @@ def main() -> None:
         elector = LeaderElector(kube, "72dd1cf1.llm-d.ai", args.configmap_namespace, identity)
 
     stop = threading.Event()
+    wake = threading.Event()  # VA-create events trigger an immediate tick
     signal.signal(signal.SIGTERM, lambda *_: stop.set())
     signal.signal(signal.SIGINT, lambda *_: stop.set())
+
+    def watch_loop():
+        """Create-only VA watch (the reference's event filter: Create -> true,
+        Update/Delete/Generic -> false, controller.go:456-487)."""
+        while not stop.is_set():
+            try:
+                for etype, _va in kube.watch_events(timeout_seconds=55):
+                    if etype == "ADDED":
+                        wake.set()
+                    if stop.is_set():
+                        return
+            except Exception:
+                stop.wait(5.0)
+
+    threading.Thread(target=watch_loop, daemon=True).start()
 
     state["ready"] = True
     while not stop.is_set():
         if elector is not None and not elector.try_acquire():
             time.sleep(2.0)
             continue
+        wake.clear()
         result = reconciler.reconcile()
         logger.info(
             "reconcile complete",
@@ -137,7 +154,11 @@ def main() -> None:
                 }
             },
         )
-        stop.wait(result.requeue_after)
+        # sleep until the requeue interval elapses or a VA-create event fires
+        deadline = time.monotonic() + result.requeue_after
+        while not stop.is_set() and time.monotonic() < deadline:
+            if wake.wait(timeout=min(1.0, max(deadline - time.monotonic(), 0.05))):
+                break
 
 
 if __name__ == "__main__":
