@@ -377,3 +377,36 @@ class TestReconcile:
         status = kube.status_updates[0]
         assert status["currentAlloc"]["load"]["arrivalRate"] == "120.00"
         assert status["desiredOptimizedAlloc"]["accelerator"] == "MI355X"
+
+
+class TestPartialInfeasible:
+    def test_infeasible_variant_skipped_others_proceed(self):
+        """One variant with an impossible SLO is skipped; the rest of the
+        fleet still gets optimized (per-VA continue-on-error)."""
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=2.0)
+        # add a second VA with an SLO below its alpha (never satisfiable)
+        bad = make_va(name="bad-deploy", model="meta/llama0-70b", acc="MI355X")
+        bad.spec.modelProfile.accelerators[0].perfParms.decodeParms = {
+            "alpha": "500.0", "beta": "1.0"  # ITL floor 500ms >> slo-tpot 80
+        }
+        kube.add_va(bad)
+        kube.add_deployment(
+            Deployment(name="bad-deploy", namespace="default", replicas=1,
+                       status_replicas=1, uid="uid-bad")
+        )
+        now = time.time()
+        for q in (
+            collector.arrival_query("meta/llama0-70b", "default"),
+            collector.avg_prompt_tokens_query("meta/llama0-70b", "default"),
+            collector.avg_decode_tokens_query("meta/llama0-70b", "default"),
+            collector.ttft_query("meta/llama0-70b", "default"),
+            collector.itl_query("meta/llama0-70b", "default"),
+        ):
+            prom.results[q] = [Sample(2.0, now)]
+        result = rec.reconcile()
+        # the good variant got a decision; the infeasible one got none
+        good = kube.vas[("default", "vllme-deploy")]
+        assert good.status.desiredOptimizedAlloc.numReplicas >= 1
+        bad_stored = kube.vas[("default", "bad-deploy")]
+        assert bad_stored.status.desiredOptimizedAlloc.accelerator == ""
+        assert result.processed == 1
