@@ -427,7 +427,7 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
 // dynamic LDS: S[0..maxN] prefix, then scratch[2*NT/64] for reductions.
 // ---------------------------------------------------------------------------
 template <int NT>
-__global__ void __launch_bounds__(NT) wva_sweep_t(WvaCellsIn in, WvaCellsOut out, int n_blocks,
+__global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut out, int n_blocks,
                                                   const int *cell_ids, int max_n) {
   extern __shared__ double smem[];
   if ((int)blockIdx.x >= n_blocks) return;
