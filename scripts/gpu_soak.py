@@ -40,7 +40,7 @@ def main() -> None:
         kw = dict(
             n_servers=args.servers,
             seed=seed,
-            arrival_scale=[6.0, 60.0, 600.0, 6000.0][f % 4],
+            arrival_scale=[0.06, 6.0, 60.0, 600.0, 6000.0, 60000.0][f % 6],
             min_num_replicas=[0, 1][f % 2],
             n_accelerators=[1, 2, 3][f % 3],
         )
