@@ -33,6 +33,14 @@ def _parse_args():
     p.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
     p.add_argument("--seed", type=int, default=1234)
     p.add_argument(
+        "--limited",
+        type=int,
+        default=0,
+        metavar="CAP",
+        help="limited mode: greedy capacity-constrained solver with CAP units "
+        "per accelerator type (0 = unlimited argmin, the default)",
+    )
+    p.add_argument(
         "--preset",
         choices=["config2", "config3", "config4", "config5"],
         default=None,
@@ -106,6 +114,13 @@ def main():
     spec, args.models_per_gpu = preset_fleet(
         args.preset, args.models_per_gpu, world, args.seed
     )
+    if args.limited > 0:
+        from inferno_amd.config import AcceleratorCount, OptimizerSpec
+
+        spec.optimizer = OptimizerSpec(unlimited=False, saturationPolicy="PriorityRoundRobin")
+        spec.capacity = [
+            AcceleratorCount(type=a.type, count=args.limited) for a in spec.accelerators
+        ]
     system, opt_spec = System.from_spec(spec)
     for acc in system.accelerators.values():
         acc.calculate()
@@ -244,9 +259,12 @@ def main():
             "reconcile_p50_ms": round(p50, 3),
             "reconcile_p95_ms": round(p95, 3),
             "cells_per_step": total_cells_per_step,
-            "solver": "unlimited argmin (HIP wva_sweep + wva_argmin)"
-            if engine.backend == "gpu"
-            else "unlimited argmin (CPU golden)",
+            "solver": (
+                ("greedy limited (PriorityRoundRobin)" if args.limited > 0
+                 else "unlimited argmin")
+                + (" (HIP wva_sweep + wva_argmin)" if engine.backend == "gpu"
+                   else " (CPU golden)")
+            ),
         },
     }
     if rank == 0:
