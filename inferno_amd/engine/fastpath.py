@@ -365,6 +365,42 @@ class FastSweep:
         st["pin_out_f"].copy_(gf, non_blocking=True)
         st["pin_out_i"].copy_(gi, non_blocking=True)
 
+    def reconcile_cells(self) -> Optional[dict]:
+        """Run the sweep and return PER-CELL output arrays (numpy) plus the
+        cell metadata — the input the greedy limited-mode solver needs (full
+        candidate lists, not just argmin winners). GPU backend only; returns
+        None on CPU (the slow engine path covers it)."""
+        import torch
+
+        from ..ops.sweep import load_library
+
+        if self.backend != "gpu" or self.n_cells == 0:
+            return None
+        if not hasattr(self, "_gpu"):
+            self._init_gpu_state()
+        st = self._gpu
+        lib = load_library(allow_build=False)
+        arrs = self._refresh_dynamic()
+        pin_i, pin_f = st["pin_i"], st["pin_f"]
+        for j, k in enumerate(self._DYN_INT):
+            pin_i[j] = torch.from_numpy(np.ascontiguousarray(arrs[k], dtype=np.int32))
+        pin_f[0] = torch.from_numpy(
+            np.ascontiguousarray(arrs["arrival_rate"], dtype=np.float32)
+        )
+        pin_f[1] = torch.from_numpy(np.ascontiguousarray(arrs["cur_cost"], dtype=np.float32))
+        buckets = self._buckets_for(arrs["batch_n"])
+        self._launch_all(lib, buckets)
+        torch.cuda.current_stream().synchronize()
+        out = {
+            k: st[k].cpu().numpy()
+            for k in ("feasible", "zero_empty", "num_replicas", "batch", "cost",
+                      "value", "itl", "ttft", "rho", "max_rate")
+        }
+        out["cell_server"] = self.cell_server
+        out["cell_acc_idx"] = self.cell_acc_idx
+        out["seg_start"] = self.seg_start
+        return out
+
     def _reconcile_gpu(self) -> WinnerRecord:
         import os
 

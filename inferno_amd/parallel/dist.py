@@ -266,8 +266,54 @@ class ShardedSolver:
         stats = _ShardStats(n_cells=self._fast_sweep.n_cells, n_servers=len(local_names))
         return rec, stats
 
+    def _populate_from_cells(self, system: System, local_names, cells) -> None:
+        """Fill server.all_allocations from per-cell GPU output arrays (only
+        feasible cells are materialized as objects)."""
+        from ..core import Allocation
+
+        acc_names = sorted(system.accelerators)
+        for name in local_names:
+            system.servers[name].all_allocations = {}
+        feasible = cells["feasible"]
+        for i in np.nonzero(feasible)[0]:
+            server = system.servers[local_names[int(cells["cell_server"][i])]]
+            acc_key = acc_names[int(cells["cell_acc_idx"][i])]
+            alloc = Allocation(
+                accelerator="" if cells["zero_empty"][i] else acc_key,
+                num_replicas=int(cells["num_replicas"][i]),
+                batch_size=int(cells["batch"][i]),
+                cost=float(cells["cost"][i]),
+                value=float(cells["value"][i]),
+                itl=float(cells["itl"][i]),
+                ttft=float(cells["ttft"][i]),
+                rho=float(cells["rho"][i]),
+                max_arrv_rate_per_replica=float(cells["max_rate"][i]),
+            )
+            server.all_allocations[acc_key] = alloc
+
     def _solve_slow(self, system: System, local_names, spec: OptimizerSpec, acc_names):
+        # greedy limited mode on GPU: candidates from the cached FastSweep
+        # (K4 plan: GPU sweep produces the sorted-candidate inputs, the
+        # sequential capacity loop stays host-side)
+        if self.engine.backend == "gpu" and not spec.unlimited:
+            key = (id(system), tuple(local_names))
+            if self._fast_key != key:
+                self._fast_sweep = FastSweep(
+                    system, local_names, backend="gpu", device=self.engine.device
+                )
+                self._fast_key = key
+            cells = self._fast_sweep.reconcile_cells()
+            if cells is not None:
+                self._populate_from_cells(system, local_names, cells)
+                stats = _ShardStats(
+                    n_cells=self._fast_sweep.n_cells, n_servers=len(local_names)
+                )
+                return self._finish_slow(system, local_names, spec, acc_names, stats)
         stats = self.engine.sweep(system, server_names=local_names)
+        return self._finish_slow(system, local_names, spec, acc_names, stats)
+
+    def _finish_slow(self, system: System, local_names, spec: OptimizerSpec, acc_names,
+                     stats):
         if spec.unlimited:
             for name in local_names:
                 server = system.servers[name]

@@ -248,3 +248,22 @@ class TestControllerOnGpu:
         va = kube.vas[("default", "vllme-deploy")]
         assert va.status.desiredOptimizedAlloc.numReplicas >= 1
         assert api.is_condition_true(va, api.TYPE_OPTIMIZATION_READY)
+
+
+class TestShardedGreedyGpu:
+    def test_sharded_limited_matches_cpu(self):
+        from inferno_amd.engine import SweepEngine
+        from inferno_amd.parallel import ShardedSolver
+
+        cap = {"AMD-MI355X-288GB": 10, "AMD-MI325X-256GB": 10, "AMD-MI300X-192GB": 10}
+        a, opt = System.from_spec(make_spec(n_servers=8, seed=600, unlimited=False,
+                                            capacity=dict(cap)))
+        b, _ = System.from_spec(make_spec(n_servers=8, seed=600, unlimited=False,
+                                          capacity=dict(cap)))
+        g = ShardedSolver(SweepEngine(backend="gpu")).solve(a, opt)
+        c = ShardedSolver(SweepEngine(backend="cpu")).solve(b, opt)
+        assert set(g.solution) == set(c.solution)
+        for name in g.solution:
+            ga, ca = g.solution[name], c.solution[name]
+            assert ga.accelerator == ca.accelerator
+            assert abs(ga.numReplicas - ca.numReplicas) <= 1
