@@ -90,6 +90,45 @@ class PrometheusClient:
         self._client.close()
 
 
+def prometheus_config_from_env(cm: Optional[dict] = None) -> dict:
+    """Prometheus client kwargs from env vars (the reference's env-name set,
+    tls.go:98-117 ParsePrometheusConfigFromEnv) with ConfigMap fallback for
+    each key (controller.go:541-580): PROMETHEUS_BASE_URL, _TLS_INSECURE_
+    SKIP_VERIFY, _CA_CERT_PATH, _CLIENT_CERT_PATH/_CLIENT_KEY_PATH,
+    _BEARER_TOKEN, _TOKEN_PATH."""
+    import os
+
+    cm = cm or {}
+
+    def get(key: str, default: str = "") -> str:
+        return os.environ.get(key) or cm.get(key, default)
+
+    token = get("PROMETHEUS_BEARER_TOKEN")
+    if not token:
+        token_path = get("PROMETHEUS_TOKEN_PATH")
+        if token_path and os.path.exists(token_path):
+            with open(token_path) as f:
+                token = f.read().strip()
+    cert = get("PROMETHEUS_CLIENT_CERT_PATH")
+    key = get("PROMETHEUS_CLIENT_KEY_PATH")
+    return {
+        "base_url": get("PROMETHEUS_BASE_URL"),
+        "token": token or None,
+        "ca_cert": get("PROMETHEUS_CA_CERT_PATH") or None,
+        "client_cert": (cert, key) if cert and key else None,
+        "insecure_skip_verify": get("PROMETHEUS_TLS_INSECURE_SKIP_VERIFY") == "true",
+        "allow_http": get("PROMETHEUS_ALLOW_HTTP") == "true",
+    }
+
+
+def validate_prometheus_api(prom: PromAPI, backoff=None) -> None:
+    """Startup connectivity check with an "up" query under exponential
+    backoff (ref internal/utils/utils.go:390-410 ValidatePrometheusAPI)."""
+    from ..utils.backoff import PROMETHEUS_BACKOFF, retry_with_backoff
+
+    retry_with_backoff(lambda: prom.query("up"), backoff or PROMETHEUS_BACKOFF)
+
+
 def fix_value(x: float) -> float:
     """NaN/Inf -> 0 (ref collector.go:281-285)."""
     if math.isnan(x) or math.isinf(x):

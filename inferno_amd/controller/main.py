@@ -90,17 +90,15 @@ def main() -> None:
     serve_probes(args.metrics_port, state)
 
     kube = HttpKube()
-    prom_url = os.environ.get("PROMETHEUS_BASE_URL", "")
-    if not prom_url:
-        cm = kube.get_configmap(args.configmap_namespace, WVA_CONFIG_CM) or {}
-        prom_url = cm.get("PROMETHEUS_BASE_URL", "")
-    prom = collector.PrometheusClient(
-        prom_url,
-        token=os.environ.get("PROMETHEUS_BEARER_TOKEN") or None,
-        ca_cert=os.environ.get("PROMETHEUS_CA_CERT_PATH") or None,
-        insecure_skip_verify=os.environ.get("PROMETHEUS_TLS_INSECURE_SKIP_VERIFY") == "true",
-        allow_http=os.environ.get("PROMETHEUS_ALLOW_HTTP") == "true",
-    )
+    cm = kube.get_configmap(args.configmap_namespace, WVA_CONFIG_CM) or {}
+    prom_cfg = collector.prometheus_config_from_env(cm)
+    prom = collector.PrometheusClient(**prom_cfg)
+    # startup connectivity validation with backoff (reference: "up" query)
+    try:
+        collector.validate_prometheus_api(prom)
+    except Exception as e:  # noqa: BLE001
+        logger.error("Prometheus validation failed after retries", extra={"kv": {"err": str(e)}})
+        raise SystemExit(1)
 
     emitter = init_metrics()
     reconciler = Reconciler(
