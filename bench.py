@@ -33,6 +33,14 @@ def _parse_args():
     p.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
     p.add_argument("--seed", type=int, default=1234)
     p.add_argument(
+        "--cpu-analyzer",
+        dest="analyzer",
+        choices=["mm1k", "mg1"],
+        default=None,
+        help="queueing evaluator: mm1k = reference state-dependent chain "
+        "(default), mg1 = closed-form M/G/1/K (BASELINE config 4)",
+    )
+    p.add_argument(
         "--limited",
         type=int,
         default=0,
@@ -63,6 +71,10 @@ def preset_fleet(preset, models_per_gpu, world, seed):
     if preset == "config3":
         accs = [accelerator_spec(MI355X, tp) for tp in (1, 2, 4, 8)][:4]
         return make_fleet_spec(64 * world, seed=seed, accelerators=accs), 64
+    if preset == "config4":
+        spec = make_fleet_spec(512 * world, seed=seed)
+        spec.optimizer.analyzer = "mg1"  # config 4 names the M/G/1 model
+        return spec, 512
     if preset == "config5":
         # 8 variants per model: the 3-SKU ladder + MI355X TP 2/4/8 + 2 spot tiers
         from dataclasses import replace
@@ -114,10 +126,17 @@ def main():
     spec, args.models_per_gpu = preset_fleet(
         args.preset, args.models_per_gpu, world, args.seed
     )
+    if args.analyzer:
+        spec.optimizer.analyzer = args.analyzer
     if args.limited > 0:
         from inferno_amd.config import AcceleratorCount, OptimizerSpec
 
-        spec.optimizer = OptimizerSpec(unlimited=False, saturationPolicy="PriorityRoundRobin")
+        spec.optimizer = OptimizerSpec(
+            unlimited=False,
+            saturationPolicy="PriorityRoundRobin",
+            analyzer=spec.optimizer.analyzer,
+            analyzerCV2=spec.optimizer.analyzerCV2,
+        )
         spec.capacity = [
             AcceleratorCount(type=a.type, count=args.limited) for a in spec.accelerators
         ]
@@ -259,6 +278,7 @@ def main():
             "reconcile_p50_ms": round(p50, 3),
             "reconcile_p95_ms": round(p95, 3),
             "cells_per_step": total_cells_per_step,
+            "analyzer": spec.optimizer.analyzer,
             "solver": (
                 ("greedy limited (PriorityRoundRobin)" if args.limited > 0
                  else "unlimited argmin")

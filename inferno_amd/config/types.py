@@ -127,11 +127,16 @@ class ServerSpec:
 
 @dataclass
 class OptimizerSpec:
-    """Ref: pkg/config/types.go:151-155."""
+    """Ref: pkg/config/types.go:151-155, extended with the evaluator choice
+    (this build): analyzer "mm1k" = the reference's state-dependent M/M/1/K
+    chain (default), "mg1" = the closed-form M/G/1/K cheap path with
+    squared-CV analyzerCV2 (BASELINE config 4)."""
 
     unlimited: bool = False
     delayedBestEffort: bool = False
     saturationPolicy: str = "None"
+    analyzer: str = "mm1k"
+    analyzerCV2: float = 1.0
 
 
 @dataclass
@@ -279,6 +284,8 @@ def system_spec_from_json(doc: dict[str, Any]) -> SystemSpec:
         unlimited=bool(opt_d.get("unlimited", False)),
         delayedBestEffort=bool(opt_d.get("delayedBestEffort", False)),
         saturationPolicy=opt_d.get("saturationPolicy", "None") or "None",
+        analyzer=opt_d.get("analyzer", "mm1k") or "mm1k",
+        analyzerCV2=float(opt_d.get("analyzerCV2", 1.0)),
     )
     capacity = [
         AcceleratorCount(type=c.get("type", ""), count=int(c.get("count", 0)))

@@ -236,7 +236,12 @@ def create_allocation(system, server_name: str, acc_name: str) -> Optional[Alloc
     )
     req = RequestSize(avg_input_tokens=load.avgInTokens, avg_output_tokens=K)
     try:
-        qa = QueueAnalyzer(cfg, req)
+        if getattr(system, "analyzer_mode", "mm1k") == "mg1":
+            from ..analyzer.mg1 import MG1QueueEvaluator
+
+            qa = MG1QueueEvaluator(cfg, req, cv2=getattr(system, "analyzer_cv2", 1.0))
+        else:
+            qa = QueueAnalyzer(cfg, req)
     except AnalyzerError:
         return None
 

@@ -201,6 +201,8 @@ class System:
     """Registry of accelerators/models/classes/servers. Reentrant (no singleton)."""
 
     def __init__(self) -> None:
+        self.analyzer_mode: str = "mm1k"  # or "mg1" (closed-form cheap path)
+        self.analyzer_cv2: float = 1.0
         self.accelerators: dict[str, Accelerator] = {}
         self.models: dict[str, Model] = {}
         self.service_classes: dict[str, ServiceClass] = {}
@@ -226,6 +228,8 @@ class System:
             system.servers[sv.name] = Server(sv)
         for c in spec.capacity:
             system.capacity[c.type] = c.count
+        system.analyzer_mode = spec.optimizer.analyzer or "mm1k"
+        system.analyzer_cv2 = spec.optimizer.analyzerCV2
         return system, spec.optimizer
 
     def calculate(self) -> None:

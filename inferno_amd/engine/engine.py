@@ -101,7 +101,9 @@ class SweepEngine:
                 system.servers[name].all_allocations = {}
             self._last_gpu = None
             return
-        out = run_sweep(snap.arrays, device=self.device)
+        mode = 1 if getattr(system, "analyzer_mode", "mm1k") == "mg1" else 0
+        out = run_sweep(snap.arrays, device=self.device, analyzer_mode=mode,
+                        cv2=getattr(system, "analyzer_cv2", 1.0))
         # keep device-side results for the argmin kernel
         self._last_gpu = (snap, out)
         # download once, build Allocation objects

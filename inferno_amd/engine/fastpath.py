@@ -319,6 +319,10 @@ class FastSweep:
                 ctypes.c_int(max(bmax, 1)),
                 ctypes.c_int(nt),
                 p(ids) if ids is not None else None,
+                ctypes.c_int(
+                    1 if getattr(self.system, "analyzer_mode", "mm1k") == "mg1" else 0
+                ),
+                ctypes.c_float(float(getattr(self.system, "analyzer_cv2", 1.0))),
                 ctypes.c_void_p(cur.cuda_stream),
                 p(di[0]), p(di[1]), p(di[2]), p(st["min_replicas"]), p(di[3]),
                 p(di[4]), p(di[5]),

@@ -300,14 +300,60 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
   return o;
 }
 
+// ---------------------------------------------------------------------------
+// analyzer mode 1: closed-form M/G/1/K (BASELINE config 4's "M/G/1 model").
+// mu = s(N) (marginal per-request rate at max batch, req/msec), K = 11N,
+// Pollaczek-Khinchine wait factor (1+cv2)/2, utilization-based effective
+// concurrency eff = rho*N. O(1) per evaluation — no chain, no LDS.
+// ---------------------------------------------------------------------------
+struct Mg1Out {
+  double throughput;  // req/msec
+  double wait;        // msec
+  double eff;         // effective concurrency
+  double rho;         // clamped utilization
+};
+
+__device__ __forceinline__ Mg1Out mg1_eval(double lam, double mu, int K, double cv2, int N) {
+  const double rho = lam / mu;
+  double p0, pK, avg_n;
+  if (fabs(rho - 1.0) < 1e-15) {
+    p0 = 1.0 / (double)(K + 1);
+    pK = p0;
+    avg_n = (double)K * 0.5;
+  } else {
+    const double rK1 = pow(rho, (double)(K + 1));
+    p0 = (1.0 - rho) / (1.0 - rK1);
+    pK = p0 * pow(rho, (double)K);
+    avg_n = rho / (1.0 - rho) - (double)(K + 1) * rK1 / (1.0 - rK1);
+  }
+  const double X = lam * (1.0 - pK);
+  const double serv = 1.0 / mu;
+  const double wait = fmax(avg_n / X - serv, 0.0) * (1.0 + cv2) * 0.5;
+  Mg1Out o;
+  o.throughput = X;
+  o.wait = wait;
+  const double rc = fmin(fmax(rho, 0.0), 1.0);
+  o.rho = rc;
+  o.eff = rc * (double)N;
+  return o;
+}
+
 template <int NT, int PART>
 __device__ double eval_metric(int kind, double lam, const ChainGeom &g, double logsN, int N, int K,
                               float gamma, float delta, float alpha, float beta, int in_tok,
-                              int out_tok, double *scratch) {
-  ChainOut c = chain_eval<NT, PART>(lam, g, logsN, N, K, scratch);
-  double eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+                              int out_tok, double *scratch, int mode, double cv2, double mu) {
+  double wait, eff;
+  if (mode == 1) {
+    Mg1Out o = mg1_eval(lam, mu, K, cv2, N);
+    wait = o.wait;
+    eff = o.eff;
+  } else {
+    ChainOut c = chain_eval<NT, PART>(lam, g, logsN, N, K, scratch);
+    wait = c.wait;
+    eff = effective_concurrency(c.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+  }
   if (kind == 0)
-    return c.wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
+    return wait + (double)prefill_time_f(gamma, delta, in_tok, (float)eff);
   return (double)decode_time_f(alpha, beta, (float)eff);
 }
 
@@ -331,7 +377,7 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
                             const ChainGeom &g, double logsN, int N, int K, float gamma,
                             float delta, float alpha, float beta, int in_tok, int out_tok,
                             double *scratch, double *res_slot, double lam_star[2],
-                            int ind_out[2]) {
+                            int ind_out[2], int mode, double cv2, double mu) {
   constexpr int W = NT / 2;
   const int tid = threadIdx.x;
   const int h = tid / W;  // 0 = TTFT, 1 = ITL
@@ -358,7 +404,7 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
       x_eval = x_max;  // done: dummy evaluation to keep barriers uniform
     const double y =
         eval_metric<NT, 2>(h, x_eval, g, logsN, N, K, gamma, delta, alpha, beta, in_tok,
-                           out_tok, scratch);
+                           out_tok, scratch, mode, cv2, mu);
     if (phase == 0) {
       y_lo = y;
       if (within_tol(y_lo, target)) {
@@ -428,7 +474,8 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
 // ---------------------------------------------------------------------------
 template <int NT>
 __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut out, int n_blocks,
-                                                  const int *cell_ids, int max_n) {
+                                                  const int *cell_ids, int max_n,
+                                                  int analyzer_mode, float cv2) {
   extern __shared__ double smem[];
   if ((int)blockIdx.x >= n_blocks) return;
   const int cell = cell_ids ? cell_ids[blockIdx.x] : (int)blockIdx.x;
@@ -513,6 +560,7 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   if (in_tok == 0 && out_tok == 1) num_decode = 1;
 
   // ---- build chain geometry: transposed 1/s + anchor log-prefix ----
+  // (skipped entirely in M/G/1 mode — its evaluations are closed-form)
   const int chunk = (N + NT - 1) / NT;
   const int ksub = (chunk + WVA_SUB - 1) / WVA_SUB;
   // LDS partition (S points at the dynamic smem base; scratch precedes)
@@ -521,7 +569,7 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   double *total_slot = scratch + 32;
 
   const int n0 = tid * chunk + 1;
-  const int n1 = min(n0 + chunk - 1, N);
+  const int n1 = (analyzer_mode == 1) ? 0 : min(n0 + chunk - 1, N);
   double local = 0.0;
   for (int n = n0; n <= n1; ++n) {
     float nf = (float)n;
@@ -575,13 +623,16 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
 
   const double lam_min = (double)s1 * WVA_EPSILON;
   const double lam_max = (double)sN * (1.0 - WVA_EPSILON);
+  const double mu = (double)sN;  // M/G/1 service rate (analyzer mode 1)
+  const double cv2d = (double)cv2;
   const int Kstates = 11 * N;  // maxQueue (10N) + N  (ref allocation.go:87)
 
   // ---- SLO sizing: concurrent TTFT+ITL bisections (queueanalyzer.go:185-255)
   double lam_star[2];
   int inds[2];
   dual_bisect<NT>(lam_min, lam_max, t_ttft, t_itl, geom, logsN, N, Kstates, gamma, delta,
-                  alpha, beta, in_tok, out_tok, scratch, total_slot + 1, lam_star, inds);
+                  alpha, beta, in_tok, out_tok, scratch, total_slot + 1, lam_star, inds,
+                  analyzer_mode, cv2d, mu);
   bool feasible = true;
   double lam_ttft = lam_max;
   if (t_ttft > 0.0f) {
@@ -604,8 +655,13 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   double lam = fmin(fmin(lam_ttft, lam_itl), lam_tps);
 
   // ---- analyze at sized rate -> rate* (ref allocation.go:126-131) ----
-  ChainOut c = chain_eval<NT, 1>(lam, geom, logsN, N, Kstates, scratch);
-  const double rate_star = c.throughput * 1000.0;  // req/sec
+  double tput_at_lam;
+  if (analyzer_mode == 1) {
+    tput_at_lam = mg1_eval(lam, mu, Kstates, cv2d, N).throughput;
+  } else {
+    tput_at_lam = chain_eval<NT, 1>(lam, geom, logsN, N, Kstates, scratch).throughput;
+  }
+  const double rate_star = tput_at_lam * 1000.0;  // req/sec
 
   double total_rate;  // req/sec (ref allocation.go:134-139)
   if (t_tps == 0.0f)
@@ -622,12 +678,20 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
 
   // ---- per-replica analyze (ref allocation.go:148-157) ----
   const double rate = total_rate / (double)num_replicas;
-  ChainOut c2 = chain_eval<NT, 1>(rate / 1000.0, geom, logsN, N, Kstates, scratch);
-  const double eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+  double wait2, eff, rho;
+  if (analyzer_mode == 1) {
+    Mg1Out o2 = mg1_eval(rate / 1000.0, mu, Kstates, cv2d, N);
+    wait2 = o2.wait;
+    eff = o2.eff;
+    rho = o2.rho;
+  } else {
+    ChainOut c2 = chain_eval<NT, 1>(rate / 1000.0, geom, logsN, N, Kstates, scratch);
+    wait2 = c2.wait;
+    eff = effective_concurrency(c2.serv, gamma, alpha, delta, beta, in_tok, out_tok, N);
+    rho = fmin(fmax(c2.in_servers / (double)N, 0.0), 1.0);
+  }
   const float prefill_t = prefill_time_f(gamma, delta, in_tok, (float)eff);
   const float token_t = decode_time_f(alpha, beta, (float)eff);
-  double rho = c2.in_servers / (double)N;
-  rho = fmin(fmax(rho, 0.0), 1.0);
 
   if (tid == 0) {
     out.feasible[cell] = 1;
@@ -636,7 +700,7 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
     out.batch[cell] = N;
     out.cost[cell] = cost;
     out.itl[cell] = token_t;
-    out.ttft[cell] = (float)c2.wait + prefill_t;
+    out.ttft[cell] = (float)wait2 + prefill_t;
     out.rho[cell] = (float)rho;
     out.max_rate[cell] = (float)(rate_star / 1000.0);
     float value = cost;
@@ -686,7 +750,8 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
 // C ABI launchers
 // ---------------------------------------------------------------------------
 extern "C" int wva_sweep_launch_bucket(
-    int n_blocks, int max_n, int nt, const int *cell_ids, void *stream,
+    int n_blocks, int max_n, int nt, const int *cell_ids, int analyzer_mode, float cv2,
+    void *stream,
     const int *in_tok, const int *out_tok, const int *batch_n, const int *min_replicas,
     const int *perf_max_batch, const int *cur_replicas, const int *flags,
     const float *alpha, const float *beta, const float *gamma, const float *delta,
@@ -709,15 +774,16 @@ extern "C" int wva_sweep_launch_bucket(
   switch (nt) {
     case 64:
       hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n);
+                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
       break;
     case 256:
       hipLaunchKernelGGL(wva_sweep_t<256>, dim3(n_blocks), dim3(256), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n);
+                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
       break;
     case 1024:
       hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(n_blocks), dim3(1024), lds,
-                         (hipStream_t)stream, in, out, n_blocks, cell_ids, max_n);
+                         (hipStream_t)stream, in, out, n_blocks, cell_ids, max_n,
+                         analyzer_mode, cv2);
       break;
     default:
       return -3;
@@ -736,7 +802,8 @@ extern "C" int wva_sweep_launch(
     uint8_t *feasible, uint8_t *zero_empty, int *num_replicas, int *batch, float *cost,
     float *value, float *itl, float *ttft, float *rho, float *max_rate) {
   int nt = (max_n <= WVA_N_SMALL) ? 64 : (max_n <= WVA_N_MED ? 256 : 1024);
-  return wva_sweep_launch_bucket(n_cells, max_n, nt, nullptr, stream, in_tok, out_tok, batch_n,
+  return wva_sweep_launch_bucket(n_cells, max_n, nt, nullptr, 0, 1.0f, stream, in_tok, out_tok,
+                                 batch_n,
                                  min_replicas, perf_max_batch, cur_replicas, flags, alpha, beta,
                                  gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
                                  cur_cost, feasible, zero_empty, num_replicas, batch, cost,

@@ -105,7 +105,8 @@ class SweepOutputs:
     max_rate: "object"
 
 
-def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
+def run_sweep(arrays: dict, device: str = "cuda", analyzer_mode: int = 0,
+              cv2: float = 1.0) -> SweepOutputs:
     """Launch the sweep kernel over cell SoA arrays (torch CPU tensors in,
     results copied back to CPU tensors).
 
@@ -162,6 +163,8 @@ def run_sweep(arrays: dict, device: str = "cuda") -> SweepOutputs:
             ctypes.c_int(max(bmax, 1)),
             ctypes.c_int(nt),
             _ptr(ids) if ids is not None else None,
+            ctypes.c_int(int(analyzer_mode)),
+            ctypes.c_float(float(cv2)),
             ctypes.c_void_p(cur.cuda_stream),
             _ptr(dev["in_tok"]),
             _ptr(dev["out_tok"]),

@@ -267,3 +267,37 @@ class TestShardedGreedyGpu:
             ga, ca = g.solution[name], c.solution[name]
             assert ga.accelerator == ca.accelerator
             assert abs(ga.numReplicas - ca.numReplicas) <= 1
+
+
+class TestMg1ModeGpu:
+    def test_mg1_gpu_matches_cpu_golden(self):
+        spec_a = make_spec(n_servers=16, seed=700)
+        spec_b = make_spec(n_servers=16, seed=700)
+        for s in (spec_a, spec_b):
+            s.optimizer.analyzer = "mg1"
+            s.optimizer.analyzerCV2 = 1.0
+        a, opt = System.from_spec(spec_a)
+        b, _ = System.from_spec(spec_b)
+        SweepEngine(backend="cpu").sweep(a)
+        SweepEngine(backend="gpu").sweep(b)
+        for name in a.servers:
+            am, bm = a.servers[name].all_allocations, b.servers[name].all_allocations
+            assert set(am) == set(bm), name
+            for acc in am:
+                assert_alloc_close(am[acc], bm[acc], name, acc)
+
+    def test_mg1_cv2_gpu(self):
+        spec_a = make_spec(n_servers=6, seed=701)
+        spec_b = make_spec(n_servers=6, seed=701)
+        for s in (spec_a, spec_b):
+            s.optimizer.analyzer = "mg1"
+            s.optimizer.analyzerCV2 = 2.5
+        a, opt = System.from_spec(spec_a)
+        b, _ = System.from_spec(spec_b)
+        SweepEngine(backend="cpu").sweep(a)
+        SweepEngine(backend="gpu").sweep(b)
+        for name in a.servers:
+            am, bm = a.servers[name].all_allocations, b.servers[name].all_allocations
+            assert set(am) == set(bm)
+            for acc in am:
+                assert_alloc_close(am[acc], bm[acc], name, acc)
