@@ -209,3 +209,95 @@ class TestApiParityExtras:
         servs = [qa.eval_serv_time(l) for l in lams]
         assert all(b >= a - 1e-12 for a, b in zip(waits, waits[1:]))
         assert all(s > 0 for s in servs)
+
+
+class TestMultiVariantStagedLoad:
+    """Multi-VA scenario with staged load increases (the reference's Kind e2e
+    test/e2e/e2e_test.go:698-1130, in process): three variants with distinct
+    SLO classes see ramping load; replica recommendations ramp accordingly
+    and independently."""
+
+    def _world(self):
+        kube = InMemoryKube()
+        kube.add_configmap(
+            NS, "accelerator-unit-costs",
+            {"MI355X": json.dumps({"device": "AMD-MI355X-288GB", "cost": "95.00"})},
+        )
+        kube.add_configmap(
+            NS, "service-classes-config",
+            {
+                "premium.yaml": (
+                    "name: Premium\npriority: 1\ndata:\n"
+                    "  - model: m/prem\n    slo-tpot: 60\n    slo-ttft: 800\n"
+                ),
+                "freemium.yaml": (
+                    "name: Freemium\npriority: 10\ndata:\n"
+                    "  - model: m/free\n    slo-tpot: 200\n    slo-ttft: 2500\n"
+                    "  - model: m/idle\n    slo-tpot: 200\n    slo-ttft: 2500\n"
+                ),
+            },
+        )
+        kube.add_configmap(NS, "workload-variant-autoscaler-variantautoscaling-config",
+                           {"GLOBAL_OPT_INTERVAL": "60s"})
+        for name, model in (("prem-deploy", "m/prem"), ("free-deploy", "m/free"),
+                            ("idle-deploy", "m/idle")):
+            va = api.VariantAutoscaling(
+                name=name, namespace="default",
+                labels={api.ACCELERATOR_LABEL: "MI355X"},
+                spec=api.VariantAutoscalingSpec(
+                    modelID=model,
+                    sloClassRef=api.ConfigMapKeyRef("service-classes-config", "x"),
+                    modelProfile=api.ModelProfile(accelerators=[
+                        api.AcceleratorProfile(
+                            acc="MI355X", accCount=1,
+                            perfParms=api.PerfParms(
+                                decodeParms={"alpha": "40.0", "beta": "0.6"},
+                                prefillParms={"gamma": "8.0", "delta": "0.02"},
+                            ),
+                            maxBatchSize=8,
+                        )
+                    ]),
+                ),
+            )
+            kube.add_va(va)
+            kube.add_deployment(Deployment(name=name, namespace="default",
+                                           replicas=1, status_replicas=1, uid=f"u-{name}"))
+        em = MetricsEmitter(registry=CollectorRegistry())
+        return kube, em
+
+    def _prom_for(self, loads: dict[str, float]):
+        now = time.time()
+        results = {}
+        for model, rps in loads.items():
+            results[collector.arrival_query(model, "default")] = [Sample(rps, now)]
+            results[collector.avg_prompt_tokens_query(model, "default")] = [Sample(64, now)]
+            results[collector.avg_decode_tokens_query(model, "default")] = [Sample(32, now)]
+            results[collector.ttft_query(model, "default")] = [Sample(0.05, now)]
+            results[collector.itl_query(model, "default")] = [Sample(0.02, now)]
+        return MockPromAPI(results=results)
+
+    def test_staged_ramp(self):
+        kube, em = self._world()
+        stages = [
+            {"m/prem": 1.0, "m/free": 1.0, "m/idle": 0.0},
+            {"m/prem": 12.0, "m/free": 2.0, "m/idle": 0.0},
+            {"m/prem": 40.0, "m/free": 12.0, "m/idle": 0.0},
+        ]
+        history = {"prem-deploy": [], "free-deploy": [], "idle-deploy": []}
+        for loads in stages:
+            rec = Reconciler(kube, self._prom_for(loads), em, backend="cpu",
+                             scale_to_zero=False)
+            result = rec.reconcile()
+            assert result.processed == 3
+            for name in history:
+                history[name].append(
+                    kube.vas[("default", name)].status.desiredOptimizedAlloc.numReplicas
+                )
+        # premium ramps fastest; freemium ramps slower; idle stays at min
+        assert history["prem-deploy"][-1] > history["prem-deploy"][0]
+        assert history["prem-deploy"][-1] >= history["free-deploy"][-1]
+        assert history["idle-deploy"] == [1, 1, 1]
+        # monotone non-decreasing under monotone load
+        for name in ("prem-deploy", "free-deploy"):
+            h = history[name]
+            assert all(b >= a for a, b in zip(h, h[1:]))
