@@ -48,7 +48,11 @@ class MG1QueueEvaluator:
         self.service_parms = cfg.service_parms
         self.request_size = req
         self.cv2 = float(cv2)
+        import numpy as np
+
         serv = build_service_rates(cfg, req)
+        if not np.all(np.isfinite(serv)) or np.any(serv <= 0):
+            raise AnalyzerError(f"invalid service rates for configuration {cfg}")
         self.mu = float(serv[-1])  # s(N), req/msec
         self.rate_min = float(serv[0]) * EPSILON * 1000.0
         self.rate_max = self.mu * (1.0 - EPSILON) * 1000.0

@@ -336,6 +336,11 @@ class QueueAnalyzer:
         self.service_parms = cfg.service_parms
         self.request_size = req
         self.serv_rate = build_service_rates(cfg, req)
+        # degenerate perf parameters (e.g. all-zero alpha/beta/gamma/delta
+        # from a malformed CR) produce non-positive or non-finite rates; the
+        # cell is infeasible rather than propagating inf/nan into the chain
+        if not np.all(np.isfinite(self.serv_rate)) or np.any(self.serv_rate <= 0):
+            raise AnalyzerError(f"invalid service rates for configuration {cfg}")
         lambda_min = float(self.serv_rate[0]) * EPSILON
         lambda_max = float(self.serv_rate[-1]) * (1.0 - EPSILON)
         self.rate_min = lambda_min * 1000.0  # req/sec

@@ -571,11 +571,13 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   const int n0 = tid * chunk + 1;
   const int n1 = (analyzer_mode == 1) ? 0 : min(n0 + chunk - 1, N);
   double local = 0.0;
+  bool bad_rate = false;
   for (int n = n0; n <= n1; ++n) {
     float nf = (float)n;
     float prefill = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok * nf);
     float decode = alpha + beta * nf;
     float s = nf / (prefill + (float)num_decode * decode);  // fp32 like the reference
+    bad_rate = bad_rate || !(s > 0.0f) || !isfinite(s);
     local += log((double)s);
     const int j = n - n0;
     inv_s_t[j * NT + tid] = (float)(1.0 / (double)s);
@@ -619,6 +621,17 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   float s1 = 1.0f / (prefill1 + (float)num_decode * (alpha + beta));
   float prefillN = (in_tok == 0) ? 0.0f : (gamma + delta * (float)in_tok * (float)N);
   float sN = (float)N / (prefillN + (float)num_decode * (alpha + beta * (float)N));
+  // degenerate perf parameters (zero/negative service times) -> infeasible
+  // cell rather than propagating inf/nan through the chain (matches the CPU
+  // golden's AnalyzerError guard)
+  const double any_bad = red_max_p<NT, 1>(bad_rate ? 1.0 : 0.0, scratch);
+  if (any_bad > 0.0 || !(s1 > 0.0f) || !(sN > 0.0f) || !isfinite(s1) || !isfinite(sN)) {
+    if (tid == 0) {
+      out.feasible[cell] = 0;
+      out.zero_empty[cell] = 0;
+    }
+    return;
+  }
   const double logsN = log((double)sN);
 
   const double lam_min = (double)s1 * WVA_EPSILON;

@@ -105,3 +105,32 @@ class TestSystemWiring:
         back = system_spec_from_json(system_spec_to_json(spec))
         assert back.optimizer.analyzer == "mg1"
         assert back.optimizer.analyzerCV2 == 1.5
+
+
+class TestDegenerateParms:
+    def test_zero_parms_infeasible(self):
+        from inferno_amd.analyzer import AnalyzerError, Configuration, DecodeParms, PrefillParms, QueueAnalyzer, RequestSize, ServiceParms
+        import pytest as _pt
+
+        cfg = Configuration(4, 40, ServiceParms(PrefillParms(0.0, 0.0), DecodeParms(0.0, 0.0)))
+        with _pt.raises(AnalyzerError):
+            QueueAnalyzer(cfg, RequestSize(10, 10))
+
+    def test_zero_parm_allocation_is_none(self):
+        spec = make_spec(n_servers=1, seed=95)
+        for m in spec.models:
+            m.decodeParms.alpha = 0.0
+            m.decodeParms.beta = 0.0
+            m.prefillParms.gamma = 0.0
+            m.prefillParms.delta = 0.0
+        system, _ = System.from_spec(spec)
+        assert create_allocation(system, "srv-0:ns", "MI355X") is None
+
+    def test_negative_beta_infeasible(self):
+        from inferno_amd.analyzer import AnalyzerError, Configuration, DecodeParms, PrefillParms, QueueAnalyzer, RequestSize, ServiceParms
+        import pytest as _pt
+
+        # decode time goes negative at high batch -> negative service rate
+        cfg = Configuration(64, 640, ServiceParms(PrefillParms(1.0, 0.001), DecodeParms(2.0, -0.5)))
+        with _pt.raises(AnalyzerError):
+            QueueAnalyzer(cfg, RequestSize(10, 10))
