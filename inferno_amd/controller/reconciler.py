@@ -88,13 +88,23 @@ class Actuator:
         if va.status.desiredOptimizedAlloc.numReplicas < 0:
             return
         current = self.current_deployment_replicas(va)
+        desired = va.status.desiredOptimizedAlloc.numReplicas
         self.emitter.emit_replica_metrics(
             va.name,
             va.namespace,
             current,
-            va.status.desiredOptimizedAlloc.numReplicas,
+            desired,
             va.status.desiredOptimizedAlloc.accelerator,
         )
+        # scaling-direction counter (the reference registers this series but
+        # never increments it — internal/metrics/metrics.go:84-101; here a
+        # recommendation delta counts as one scaling operation)
+        if desired > current:
+            self.emitter.emit_replica_scaling(va.name, va.namespace, "up",
+                                              "slo_optimization")
+        elif desired < current:
+            self.emitter.emit_replica_scaling(va.name, va.namespace, "down",
+                                              "cost_optimization")
 
 
 class Reconciler:

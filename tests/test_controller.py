@@ -410,3 +410,18 @@ class TestPartialInfeasible:
         bad_stored = kube.vas[("default", "bad-deploy")]
         assert bad_stored.status.desiredOptimizedAlloc.accelerator == ""
         assert result.processed == 1
+
+
+class TestScalingCounter:
+    def test_direction_counter_emitted_on_scale_out(self):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=50.0)
+        rec.reconcile()
+        va = kube.vas[("default", "vllme-deploy")]
+        desired = va.status.desiredOptimizedAlloc.numReplicas
+        assert desired > 1  # scale-out scenario
+        val = reg.get_sample_value(
+            "inferno_replica_scaling_total",
+            {"variant_name": "vllme-deploy", "namespace": "default",
+             "direction": "up", "reason": "slo_optimization"},
+        )
+        assert val == 1.0
