@@ -212,6 +212,30 @@ class HttpKube:
                     continue
                 yield etype, api.va_from_json(obj)
 
+    def watch_configmap_events(self, namespace: str, names: set[str],
+                               timeout_seconds: int = 60):
+        """Stream watch events for the controller's ConfigMaps (the
+        reference's Watches(ConfigMap) registration, controller.go:456-487).
+        Yields (event_type, name)."""
+        import json as _json
+
+        params = {"watch": "1", "timeoutSeconds": str(timeout_seconds)}
+        with self._client.stream(
+            "GET", f"/api/v1/namespaces/{namespace}/configmaps", params=params,
+            timeout=timeout_seconds + 10,
+        ) as r:
+            r.raise_for_status()
+            for line in r.iter_lines():
+                if not line:
+                    continue
+                try:
+                    evt = _json.loads(line)
+                except ValueError:
+                    continue
+                name = ((evt.get("object") or {}).get("metadata") or {}).get("name", "")
+                if name in names:
+                    yield evt.get("type", ""), name
+
     def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
         path = f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/{api.PLURAL}/{va.name}"
         ref = {

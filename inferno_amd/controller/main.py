@@ -132,7 +132,25 @@ def main() -> None:
             except Exception:
                 stop.wait(5.0)
 
+    def cm_watch_loop():
+        """Watch the controller ConfigMaps; any change triggers a tick
+        (controller.go:456-487 Watches(ConfigMap))."""
+        from .reconciler import ACCELERATOR_CM, SERVICE_CLASS_CM, WVA_CONFIG_CM
+
+        names = {ACCELERATOR_CM, SERVICE_CLASS_CM, WVA_CONFIG_CM}
+        while not stop.is_set():
+            try:
+                for _etype, _name in kube.watch_configmap_events(
+                    args.configmap_namespace, names, timeout_seconds=55
+                ):
+                    wake.set()
+                    if stop.is_set():
+                        return
+            except Exception:
+                stop.wait(5.0)
+
     threading.Thread(target=watch_loop, daemon=True).start()
+    threading.Thread(target=cm_watch_loop, daemon=True).start()
 
     state["ready"] = True
     while not stop.is_set():

@@ -221,3 +221,25 @@ class TestPrometheusClient:
             monkeypatch.delenv(k, raising=False)
         cfg = prometheus_config_from_env({"PROMETHEUS_BASE_URL": "https://cm:9090"})
         assert cfg["base_url"] == "https://cm:9090"
+
+
+class TestConfigMapWatch:
+    def test_filters_to_controller_configmaps(self):
+        lines = [
+            json.dumps({"type": "MODIFIED", "object": {
+                "metadata": {"name": "service-classes-config"}}}),
+            json.dumps({"type": "MODIFIED", "object": {
+                "metadata": {"name": "unrelated-cm"}}}),
+            json.dumps({"type": "ADDED", "object": {
+                "metadata": {"name": "accelerator-unit-costs"}}}),
+        ]
+
+        def handler(req):
+            assert req.url.path == "/api/v1/namespaces/ns/configmaps"
+            return httpx.Response(200, text="\n".join(lines))
+
+        events = list(make_kube(handler).watch_configmap_events(
+            "ns", {"service-classes-config", "accelerator-unit-costs"},
+            timeout_seconds=1))
+        assert events == [("MODIFIED", "service-classes-config"),
+                          ("ADDED", "accelerator-unit-costs")]
