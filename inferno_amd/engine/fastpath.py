@@ -109,7 +109,6 @@ class FastSweep:
                            "t_itl", "t_ttft", "t_tps") else np.int32)
             for k, v in stat.items()
         }
-        self._torch_cache = None
 
     # ------------------------------------------------------------------
     # Array overrides for the steady-state loop: when set, the per-server

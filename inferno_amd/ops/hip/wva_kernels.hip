@@ -6,33 +6,48 @@
 // mm1modelstatedependent.go:70-116 computeProbabilities and
 // pkg/solver/solver.go:63-79 SolveUnlimited) with batched device kernels:
 //
-//   K1 wva_sweep<NT>: one block per (server, accelerator[, TP]) cell,
-//       templated on block size and dispatched per N-bucket by the host:
-//         N <=  512  -> NT=64   (one wave; all reductions are barrier-free
-//                                __shfl_xor butterflies, scan carries in
-//                                registers — zero __syncthreads)
-//         N <= 2048  -> NT=256  (4 waves, hierarchical shfl+LDS reductions)
-//         N >  2048  -> NT=1024 (16 waves — the straggler cells' latency
-//                                sets the sweep's wall time, since ~all
-//                                cells are resident across 256 CUs at once)
-//       Per cell: fp32 state-dependent service rates (the reference's
-//       float32 inputs), fp64 log-prefix in LDS, then the TTFT and ITL SLO
-//       bisections as ~10^2 lock-step chain evaluations; each evaluation is
-//       an O(N/NT) strided pass + reductions, with the 10N saturated queue
-//       states folded into an analytic geometric tail (vs the reference's
-//       O(11N) sequential recurrence with overflow rescaling).
+//   K1 wva_sweep_t<NT>: one block per (server, accelerator[, TP]) cell,
+//       templated on block size. The host dispatches cells to 64/256/1024-
+//       thread blocks by batch size N AND by regime (ops/sweep.py
+//       choose_buckets): wide blocks shorten a single cell's latency and are
+//       used only when the bucket is small enough to be latency-bound; in
+//       the throughput regime (cells >> CUs) narrower blocks win residency.
+//       The buckets launch on separate HIP streams so they overlap.
+//
+//       Per cell (state-dependent M/M/1/K mode, the reference's evaluator):
+//       fp32 service rates s(n) (matching the reference's float32 inputs),
+//       then a chunked scan builds the chain geometry in LDS — a chunk-
+//       TRANSPOSED fp32 1/s(n) table plus the fp64 log-prefix at every
+//       32-state anchor. Each chain evaluation is O(N/NT) per lane: within a
+//       32-state sub-chunk probabilities advance by the linear-space
+//       recurrence w(n+1) = w(n)*lam/s(n+1) (ONE fp64 exp per anchor instead
+//       of per state), the 10N saturated queue states fold into an analytic
+//       geometric tail (vs the reference's O(11N) sequential recurrence with
+//       overflow rescaling), and reductions are __shfl_xor butterflies
+//       (hierarchical shfl+LDS above one wave; a 64-thread block runs
+//       entirely barrier-free). The TTFT and ITL SLO bisections
+//       (pkg/analyzer/utils.go:26-70 semantics: 1e-6 relative tolerance,
+//       <=100 iterations, below/within/above indicators) run CONCURRENTLY in
+//       the two half-blocks in lock-step (dual_bisect), halving the sizing
+//       phase's sequential depth.
+//
+//       analyzer_mode 1 selects the closed-form M/G/1/K evaluator
+//       (mg1_eval: O(1) per evaluation, no chain/LDS — BASELINE config 4).
+//
 //   K2 wva_argmin: segmented argmin over the sweep output per server
 //       (value = transition-penalty-adjusted cost), deterministic
 //       lowest-cell-index tie-break.
 //
 // Built standalone with hipcc (no torch headers); exposed as a C ABI and
-// driven from Python via ctypes on torch tensors' device pointers.
+// driven from Python via ctypes on torch tensors' device pointers
+// (ops/sweep.py one-shot API; engine/fastpath.py persistent-state hot path).
 #include <hip/hip_runtime.h>
 #include <math.h>
 #include <stdint.h>
 
 #define WVA_WAVE 64
-// max supported batch size (LDS: (N+1) doubles for the log prefix)
+// max supported batch size per cell (bounds the LDS chain geometry:
+// chunk*NT fp32 reciprocals + NT*ksub fp64 anchors + a 40-double header)
 #define WVA_MAX_N 8192
 // N-bucket thresholds (host mirrors these in ops/sweep.py)
 #define WVA_N_SMALL 512
