@@ -284,125 +284,87 @@ class FastSweep:
             st["side_streams"].append(torch.cuda.Stream())
         return buckets
 
-    def _launch_all(self, lib, buckets) -> None:
-        """The full device-side reconcile: dynamic H2D copies, the bucketed
-        sweep launches (side streams), argmin, on-device winner gather and
-        the pinned D2H copies. Capturable as ONE hipGraph — the launch
-        sequence is static for a fixed bucket layout, so steady-state
-        reconciles replay the graph instead of re-issuing ~15 launches."""
+    def _ensure_ctx(self) -> None:
+        """Create the native reconcile context (opaque C struct owning the
+        HIP streams/events and all registered device/pinned pointers) —
+        the whole per-tick pipeline then runs behind ONE C call."""
         import ctypes
-
-        import torch
-
-        from ..ops.sweep import HipKernelError
-
-        st = self._gpu
-        n_srv = len(self.server_names)
-        st["dev_i"].copy_(st["pin_i"], non_blocking=True)
-        st["dev_f"].copy_(st["pin_f"], non_blocking=True)
-
-        main_stream = torch.cuda.current_stream()
-        # ctypes argument tuples are static per bucket layout (device buffers
-        # are persistent) — build once, reuse every reconcile
-        if st.get("args_key") != st["bucket_key"]:
-            di, df = st["dev_i"], st["dev_f"]
-
-            def p(t):
-                return ctypes.c_void_p(t.data_ptr())
-
-            mode = ctypes.c_int(
-                1 if getattr(self.system, "analyzer_mode", "mm1k") == "mg1" else 0
-            )
-            cv2 = ctypes.c_float(float(getattr(self.system, "analyzer_cv2", 1.0)))
-            common = (
-                p(di[0]), p(di[1]), p(di[2]), p(st["min_replicas"]), p(di[3]),
-                p(di[4]), p(di[5]),
-                p(st["alpha"]), p(st["beta"]), p(st["gamma"]), p(st["delta"]),
-                p(df[0]),
-                p(st["t_itl"]), p(st["t_ttft"]), p(st["t_tps"]), p(st["acc_cost"]),
-                p(df[1]),
-                p(st["feasible"]), p(st["zero_empty"]), p(st["num_replicas"]),
-                p(st["batch"]), p(st["cost"]), p(st["value"]), p(st["itl"]),
-                p(st["ttft"]), p(st["rho"]), p(st["max_rate"]),
-            )
-            st["bucket_args"] = [
-                (
-                    nt,
-                    (
-                        ctypes.c_int(self.n_cells if ids is None else int(ids.numel())),
-                        ctypes.c_int(max(bmax, 1)),
-                        ctypes.c_int(nt),
-                        ctypes.c_void_p(ids.data_ptr()) if ids is not None else None,
-                        mode,
-                        cv2,
-                    ),
-                    common,
-                )
-                for nt, ids, bmax, _count in buckets
-            ]
-            st["argmin_args"] = (
-                ctypes.c_int(n_srv),
-                ctypes.c_void_p(main_stream.cuda_stream),
-                p(st["value"]), p(st["feasible"]), p(st["seg"]), p(st["winner"]),
-            )
-            st["args_key"] = st["bucket_key"]
-
-        for i, (nt, head, common) in enumerate(st["bucket_args"]):
-            if i == 0:
-                cur = main_stream
-            else:
-                cur = st["side_streams"][i - 1]
-                cur.wait_stream(main_stream)
-            rc = lib.wva_sweep_launch_bucket(
-                *head, ctypes.c_void_p(cur.cuda_stream), *common
-            )
-            if rc != 0:
-                raise HipKernelError(f"wva_sweep_launch_bucket(nt={nt}) failed: {rc}")
-        for s in st["side_streams"][: max(len(st["bucket_args"]) - 1, 0)]:
-            main_stream.wait_stream(s)
-
-        rc = lib.wva_argmin_launch(*st["argmin_args"])
-        if rc != 0:
-            raise HipKernelError(f"wva_argmin_launch failed: {rc}")
-
-        # gather winner fields on-device, pinned D2H
-        import torch as t
-
-        w = st["winner"]
-        has = w >= 0
-        wc = t.where(has, w, t.zeros_like(w)).long()
-        gf, gi = st["gather_f"], st["gather_i"]
-        zero_f = t.zeros((), dtype=t.float32, device=w.device)
-        for j, k in enumerate(("cost", "value", "itl", "ttft", "rho", "max_rate")):
-            gf[j] = t.where(has, st[k][wc], zero_f)
-        acc = t.where(
-            st["zero_empty"][wc] > 0,
-            t.full_like(w, -2),
-            st["cell_acc"][wc],
-        )
-        gi[0] = t.where(has, acc, t.full_like(w, -1))
-        gi[1] = t.where(has, st["num_replicas"][wc], t.zeros_like(w))
-        gi[2] = t.where(has, st["batch"][wc], t.zeros_like(w))
-        gi[3] = w
-        st["pin_out_f"].copy_(gf, non_blocking=True)
-        st["pin_out_i"].copy_(gi, non_blocking=True)
-
-    def reconcile_cells(self) -> Optional[dict]:
-        """Run the sweep and return PER-CELL output arrays (numpy) plus the
-        cell metadata — the input the greedy limited-mode solver needs (full
-        candidate lists, not just argmin winners). GPU backend only; returns
-        None on CPU (the slow engine path covers it)."""
-        import torch
 
         from ..ops.sweep import load_library
 
-        if self.backend != "gpu" or self.n_cells == 0:
-            return None
-        if not hasattr(self, "_gpu"):
-            self._init_gpu_state()
         st = self._gpu
+        if "ctx" in st:
+            return
         lib = load_library(allow_build=False)
-        arrs = self._refresh_dynamic()
+        di, df = st["dev_i"], st["dev_f"]
+        slots = [
+            di, df,
+            st["min_replicas"],
+            st["alpha"], st["beta"], st["gamma"], st["delta"],
+            st["t_itl"], st["t_ttft"], st["t_tps"], st["acc_cost"],
+            st["feasible"], st["zero_empty"], st["num_replicas"], st["batch"],
+            st["cost"], st["value"], st["itl"], st["ttft"], st["rho"],
+            st["max_rate"],
+            st["seg"], st["winner"], st["cell_acc"], st["gather_f"],
+            st["gather_i"],
+            st["pin_i"], st["pin_f"], st["pin_out_f"], st["pin_out_i"],
+        ]
+        arr = (ctypes.c_void_p * len(slots))(
+            *[ctypes.c_void_p(t.data_ptr()) for t in slots]
+        )
+        ctx = lib.wva_ctx_create(
+            ctypes.c_int(self.n_cells), ctypes.c_int(len(self.server_names)), arr
+        )
+        if not ctx:
+            from ..ops.sweep import HipKernelError
+
+            raise HipKernelError("wva_ctx_create failed")
+        st["ctx"] = ctx
+        st["ctx_lib"] = lib
+
+    def __del__(self):  # release the native context's streams/events
+        st = getattr(self, "_gpu", None)
+        if st and "ctx" in st:
+            try:
+                st["ctx_lib"].wva_ctx_destroy(st.pop("ctx"))
+            except Exception:
+                pass
+
+    def _sync_buckets(self, batch_n: np.ndarray) -> None:
+        """Push the bucket layout + analyzer mode into the native context
+        (only when it changed)."""
+        import ctypes
+
+        st = self._gpu
+        buckets = self._buckets_for(batch_n)
+        if st.get("ctx_buckets_key") == st["bucket_key"]:
+            return
+        n = len(buckets)
+        nts = (ctypes.c_int * n)(*[b[0] for b in buckets])
+        ids = (ctypes.c_void_p * n)(
+            *[ctypes.c_void_p(b[1].data_ptr()) if b[1] is not None else None
+              for b in buckets]
+        )
+        nbl = (ctypes.c_int * n)(
+            *[int(b[1].numel()) if b[1] is not None else self.n_cells for b in buckets]
+        )
+        mxn = (ctypes.c_int * n)(*[b[2] for b in buckets])
+        mode = 1 if getattr(self.system, "analyzer_mode", "mm1k") == "mg1" else 0
+        cv2 = float(getattr(self.system, "analyzer_cv2", 1.0))
+        rc = st["ctx_lib"].wva_ctx_set_buckets(
+            st["ctx"], ctypes.c_int(n), nts, ids, nbl, mxn,
+            ctypes.c_int(mode), ctypes.c_float(cv2),
+        )
+        if rc != 0:
+            from ..ops.sweep import HipKernelError
+
+            raise HipKernelError(f"wva_ctx_set_buckets failed: {rc}")
+        st["ctx_buckets_key"] = st["bucket_key"]
+
+    def _fill_pinned(self, arrs: dict) -> None:
+        import torch
+
+        st = self._gpu
         pin_i, pin_f = st["pin_i"], st["pin_f"]
         for j, k in enumerate(self._DYN_INT):
             pin_i[j] = torch.from_numpy(np.ascontiguousarray(arrs[k], dtype=np.int32))
@@ -410,9 +372,30 @@ class FastSweep:
             np.ascontiguousarray(arrs["arrival_rate"], dtype=np.float32)
         )
         pin_f[1] = torch.from_numpy(np.ascontiguousarray(arrs["cur_cost"], dtype=np.float32))
-        buckets = self._buckets_for(arrs["batch_n"])
-        self._launch_all(lib, buckets)
-        torch.cuda.current_stream().synchronize()
+
+    def _native_reconcile(self, arrs: dict) -> None:
+        from ..ops.sweep import HipKernelError
+
+        st = self._gpu
+        self._ensure_ctx()
+        self._fill_pinned(arrs)
+        self._sync_buckets(arrs["batch_n"])
+        rc = st["ctx_lib"].wva_reconcile(st["ctx"])
+        if rc != 0:
+            raise HipKernelError(f"wva_reconcile failed: {rc}")
+
+    def reconcile_cells(self) -> Optional[dict]:
+        """Run the sweep and return PER-CELL output arrays (numpy) plus the
+        cell metadata — the input the greedy limited-mode solver needs (full
+        candidate lists, not just argmin winners). GPU backend only; returns
+        None on CPU (the slow engine path covers it)."""
+        if self.backend != "gpu" or self.n_cells == 0:
+            return None
+        if not hasattr(self, "_gpu"):
+            self._init_gpu_state()
+        st = self._gpu
+        arrs = self._refresh_dynamic()
+        self._native_reconcile(arrs)  # fully synced on return
         out = {
             k: st[k].cpu().numpy()
             for k in ("feasible", "zero_empty", "num_replicas", "batch", "cost",
@@ -424,57 +407,14 @@ class FastSweep:
         return out
 
     def _reconcile_gpu(self) -> WinnerRecord:
-        import os
-
-        import torch
-
-        from ..ops.sweep import load_library
-
         n_srv = len(self.server_names)
         if self.n_cells == 0:
             return _empty_winner(n_srv)
         if not hasattr(self, "_gpu"):
             self._init_gpu_state()
         st = self._gpu
-        lib = load_library(allow_build=False)
         arrs = self._refresh_dynamic()
-
-        # pack the dynamic inputs into the pinned staging buffers (the graph
-        # replays the H2D copies from these fixed host pointers)
-        pin_i, pin_f = st["pin_i"], st["pin_f"]
-        for j, k in enumerate(self._DYN_INT):
-            pin_i[j] = torch.from_numpy(np.ascontiguousarray(arrs[k], dtype=np.int32))
-        pin_f[0] = torch.from_numpy(
-            np.ascontiguousarray(arrs["arrival_rate"], dtype=np.float32)
-        )
-        pin_f[1] = torch.from_numpy(np.ascontiguousarray(arrs["cur_cost"], dtype=np.float32))
-
-        buckets = self._buckets_for(arrs["batch_n"])
-        use_graph = os.environ.get("INFERNO_HIPGRAPH", "0") == "1"  # measured: parity with eager (the launch path is already ~5 launches on pipelined streams); opt-in
-        if use_graph and st.get("graph_key") == st["bucket_key"]:
-            st["graph"].replay()
-        else:
-            captured = False
-            if use_graph:
-                try:
-                    # warm-up pass (allocator state), then capture the whole
-                    # launch sequence as a hipGraph
-                    self._launch_all(lib, buckets)
-                    torch.cuda.synchronize()
-                    graph = torch.cuda.CUDAGraph()
-                    with torch.cuda.graph(graph):
-                        self._launch_all(lib, buckets)
-                    st["graph"] = graph
-                    st["graph_key"] = st["bucket_key"]
-                    graph.replay()
-                    captured = True
-                except Exception:
-                    st.pop("graph", None)
-                    st["graph_key"] = None
-                    st["use_graph_failed"] = True
-            if not captured:
-                self._launch_all(lib, buckets)
-        torch.cuda.current_stream().synchronize()
+        self._native_reconcile(arrs)
         of = st["pin_out_f"].numpy()
         oi = st["pin_out_i"].numpy()
         return WinnerRecord(

@@ -847,3 +847,241 @@ extern "C" int wva_argmin_launch(int n_servers, void *stream, const float *value
 }
 
 extern "C" int wva_device_count(int *count) { return (int)hipGetDeviceCount(count); }
+
+// ---------------------------------------------------------------------------
+// K3 wva_gather: materialize per-server winner records on-device (one thread
+// per server) so the host reads back two small pinned buffers.
+// gf rows: cost, value, itl, ttft, rho, max_rate; gi rows: acc_code
+// (-1 none, -2 zero-load empty, else accelerator index), num_replicas,
+// batch, winner cell index.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256) wva_gather(
+    const int *winner, const uint8_t *zero_empty, const int *cell_acc,
+    const int *num_replicas, const int *batch, const float *cost, const float *value,
+    const float *itl, const float *ttft, const float *rho, const float *max_rate,
+    int n_srv, float *gf, int *gi) {
+  const int s = blockIdx.x * blockDim.x + threadIdx.x;
+  if (s >= n_srv) return;
+  const int w = winner[s];
+  const bool has = w >= 0;
+  const int wc = has ? w : 0;
+  gf[0 * n_srv + s] = has ? cost[wc] : 0.0f;
+  gf[1 * n_srv + s] = has ? value[wc] : 0.0f;
+  gf[2 * n_srv + s] = has ? itl[wc] : 0.0f;
+  gf[3 * n_srv + s] = has ? ttft[wc] : 0.0f;
+  gf[4 * n_srv + s] = has ? rho[wc] : 0.0f;
+  gf[5 * n_srv + s] = has ? max_rate[wc] : 0.0f;
+  gi[0 * n_srv + s] = has ? (zero_empty[wc] ? -2 : cell_acc[wc]) : -1;
+  gi[1 * n_srv + s] = has ? num_replicas[wc] : 0;
+  gi[2 * n_srv + s] = has ? batch[wc] : 0;
+  gi[3 * n_srv + s] = w;
+}
+
+// ---------------------------------------------------------------------------
+// Native reconcile context: the whole per-tick device pipeline behind ONE C
+// call (H2D copies, regime-bucketed sweep launches overlapped via events on
+// three streams, argmin, gather, D2H, sync) — no per-step host dispatch
+// beyond wva_reconcile(ctx).
+// ---------------------------------------------------------------------------
+#define WVA_MAX_BUCKETS 3
+
+struct WvaBucket {
+  int nt;
+  const int *cell_ids;  // nullptr = identity
+  int n_blocks;
+  int max_n;
+};
+
+struct WvaCtx {
+  int n_cells;
+  int n_srv;
+  int analyzer_mode;
+  float cv2;
+  // device SoA
+  WvaCellsIn in;
+  WvaCellsOut out;
+  const int *seg_start;
+  int *winner;
+  const int *cell_acc;
+  float *gather_f;
+  int *gather_i;
+  // device dynamic blocks + host pinned sources
+  void *dev_i;        // 6 x n_cells int32
+  void *dev_f;        // 2 x n_cells f32
+  const void *pin_i;  // host pinned
+  const void *pin_f;
+  void *pin_out_f;    // host pinned, 6 x n_srv f32
+  void *pin_out_i;    // host pinned, 4 x n_srv i32
+  WvaBucket buckets[WVA_MAX_BUCKETS];
+  int n_buckets;
+  hipStream_t s0, s1, s2;
+  hipEvent_t e_up, e_b1, e_b2;
+};
+
+// pointer-slot order for wva_ctx_create (host mirrors in ops/sweep.py):
+//  0 dev_i (6*n_cells i32)   1 dev_f (2*n_cells f32)
+//  2 min_replicas            3 alpha  4 beta  5 gamma  6 delta
+//  7 t_itl  8 t_ttft  9 t_tps  10 acc_cost
+//  11 feasible  12 zero_empty  13 num_replicas  14 batch  15 cost
+//  16 value  17 itl  18 ttft  19 rho  20 max_rate
+//  21 seg_start  22 winner  23 cell_acc  24 gather_f  25 gather_i
+//  26 pin_i  27 pin_f  28 pin_out_f  29 pin_out_i
+#define WVA_CTX_SLOTS 30
+
+extern "C" void *wva_ctx_create(int n_cells, int n_srv, void **p) {
+  WvaCtx *c = new WvaCtx();
+  c->n_cells = n_cells;
+  c->n_srv = n_srv;
+  c->analyzer_mode = 0;
+  c->cv2 = 1.0f;
+  c->n_buckets = 0;
+  char *di = (char *)p[0];
+  char *df = (char *)p[1];
+  size_t ci = (size_t)n_cells * sizeof(int);
+  size_t cf = (size_t)n_cells * sizeof(float);
+  c->in.in_tok = (const int *)(di + 0 * ci);
+  c->in.out_tok = (const int *)(di + 1 * ci);
+  c->in.batch_n = (const int *)(di + 2 * ci);
+  c->in.perf_max_batch = (const int *)(di + 3 * ci);
+  c->in.cur_replicas = (const int *)(di + 4 * ci);
+  c->in.flags = (const int *)(di + 5 * ci);
+  c->in.min_replicas = (const int *)p[2];
+  c->in.alpha = (const float *)p[3];
+  c->in.beta = (const float *)p[4];
+  c->in.gamma = (const float *)p[5];
+  c->in.delta = (const float *)p[6];
+  c->in.arrival_rate = (const float *)(df + 0 * cf);
+  c->in.cur_cost = (const float *)(df + 1 * cf);
+  c->in.t_itl = (const float *)p[7];
+  c->in.t_ttft = (const float *)p[8];
+  c->in.t_tps = (const float *)p[9];
+  c->in.acc_cost = (const float *)p[10];
+  c->out.feasible = (uint8_t *)p[11];
+  c->out.zero_empty = (uint8_t *)p[12];
+  c->out.num_replicas = (int *)p[13];
+  c->out.batch = (int *)p[14];
+  c->out.cost = (float *)p[15];
+  c->out.value = (float *)p[16];
+  c->out.itl = (float *)p[17];
+  c->out.ttft = (float *)p[18];
+  c->out.rho = (float *)p[19];
+  c->out.max_rate = (float *)p[20];
+  c->seg_start = (const int *)p[21];
+  c->winner = (int *)p[22];
+  c->cell_acc = (const int *)p[23];
+  c->gather_f = (float *)p[24];
+  c->gather_i = (int *)p[25];
+  c->dev_i = p[0];
+  c->dev_f = p[1];
+  c->pin_i = p[26];
+  c->pin_f = p[27];
+  c->pin_out_f = p[28];
+  c->pin_out_i = p[29];
+  if (hipStreamCreateWithFlags(&c->s0, hipStreamNonBlocking) != hipSuccess ||
+      hipStreamCreateWithFlags(&c->s1, hipStreamNonBlocking) != hipSuccess ||
+      hipStreamCreateWithFlags(&c->s2, hipStreamNonBlocking) != hipSuccess ||
+      hipEventCreateWithFlags(&c->e_up, hipEventDisableTiming) != hipSuccess ||
+      hipEventCreateWithFlags(&c->e_b1, hipEventDisableTiming) != hipSuccess ||
+      hipEventCreateWithFlags(&c->e_b2, hipEventDisableTiming) != hipSuccess) {
+    delete c;
+    return nullptr;
+  }
+  return c;
+}
+
+extern "C" int wva_ctx_set_buckets(void *ctx, int n_buckets, const int *nts,
+                                   const void **cell_ids, const int *n_blocks,
+                                   const int *max_ns, int analyzer_mode, float cv2) {
+  WvaCtx *c = (WvaCtx *)ctx;
+  if (n_buckets < 0 || n_buckets > WVA_MAX_BUCKETS) return -1;
+  c->n_buckets = n_buckets;
+  for (int i = 0; i < n_buckets; ++i) {
+    c->buckets[i].nt = nts[i];
+    c->buckets[i].cell_ids = (const int *)cell_ids[i];
+    c->buckets[i].n_blocks = n_blocks[i];
+    c->buckets[i].max_n = max_ns[i];
+  }
+  c->analyzer_mode = analyzer_mode;
+  c->cv2 = cv2;
+  return 0;
+}
+
+static int wva_launch_bucket_on(WvaCtx *c, const WvaBucket &b, hipStream_t s) {
+  const int chunk = (b.max_n + b.nt - 1) / b.nt;
+  const int ksub = (chunk + 32 - 1) / 32;
+  size_t lds = (size_t)(40 + (chunk * b.nt + 1) / 2 + (size_t)b.nt * ksub) * sizeof(double);
+  switch (b.nt) {
+    case 64:
+      hipLaunchKernelGGL(wva_sweep_t<64>, dim3(b.n_blocks), dim3(64), lds, s, c->in, c->out,
+                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
+      break;
+    case 256:
+      hipLaunchKernelGGL(wva_sweep_t<256>, dim3(b.n_blocks), dim3(256), lds, s, c->in, c->out,
+                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
+      break;
+    case 1024:
+      hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(b.n_blocks), dim3(1024), lds, s, c->in,
+                         c->out, b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
+      break;
+    default:
+      return -3;
+  }
+  return (int)hipGetLastError();
+}
+
+extern "C" int wva_reconcile(void *ctx) {
+  WvaCtx *c = (WvaCtx *)ctx;
+  hipError_t err;
+  // 1. dynamic H2D (pinned -> device)
+  err = hipMemcpyAsync(c->dev_i, c->pin_i, (size_t)6 * c->n_cells * sizeof(int),
+                       hipMemcpyHostToDevice, c->s0);
+  if (err != hipSuccess) return (int)err;
+  err = hipMemcpyAsync(c->dev_f, c->pin_f, (size_t)2 * c->n_cells * sizeof(float),
+                       hipMemcpyHostToDevice, c->s0);
+  if (err != hipSuccess) return (int)err;
+  if (hipEventRecord(c->e_up, c->s0) != hipSuccess) return -10;
+
+  // 2. bucket launches: bucket 0 on s0; others overlap on s1/s2 after upload
+  hipStream_t side[2] = {c->s1, c->s2};
+  for (int i = 0; i < c->n_buckets; ++i) {
+    hipStream_t s = (i == 0) ? c->s0 : side[i - 1];
+    if (i > 0 && hipStreamWaitEvent(s, c->e_up, 0) != hipSuccess) return -11;
+    int rc = wva_launch_bucket_on(c, c->buckets[i], s);
+    if (rc != 0) return rc;
+    if (i == 1 && hipEventRecord(c->e_b1, s) != hipSuccess) return -12;
+    if (i == 2 && hipEventRecord(c->e_b2, s) != hipSuccess) return -13;
+  }
+  if (c->n_buckets > 1 && hipStreamWaitEvent(c->s0, c->e_b1, 0) != hipSuccess) return -14;
+  if (c->n_buckets > 2 && hipStreamWaitEvent(c->s0, c->e_b2, 0) != hipSuccess) return -15;
+
+  // 3. argmin + gather on s0
+  hipLaunchKernelGGL(wva_argmin, dim3(c->n_srv), dim3(WVA_WAVE), 0, c->s0, c->out.value,
+                     c->out.feasible, c->seg_start, c->n_srv, c->winner);
+  hipLaunchKernelGGL(wva_gather, dim3((c->n_srv + 255) / 256), dim3(256), 0, c->s0, c->winner,
+                     c->out.zero_empty, c->cell_acc, c->out.num_replicas, c->out.batch,
+                     c->out.cost, c->out.value, c->out.itl, c->out.ttft, c->out.rho,
+                     c->out.max_rate, c->n_srv, c->gather_f, c->gather_i);
+  err = hipGetLastError();
+  if (err != hipSuccess) return (int)err;
+
+  // 4. D2H of the winner records + full sync
+  err = hipMemcpyAsync(c->pin_out_f, c->gather_f, (size_t)6 * c->n_srv * sizeof(float),
+                       hipMemcpyDeviceToHost, c->s0);
+  if (err != hipSuccess) return (int)err;
+  err = hipMemcpyAsync(c->pin_out_i, c->gather_i, (size_t)4 * c->n_srv * sizeof(int),
+                       hipMemcpyDeviceToHost, c->s0);
+  if (err != hipSuccess) return (int)err;
+  return (int)hipStreamSynchronize(c->s0);
+}
+
+extern "C" void wva_ctx_destroy(void *ctx) {
+  WvaCtx *c = (WvaCtx *)ctx;
+  if (c == nullptr) return;
+  hipStreamDestroy(c->s0);
+  hipStreamDestroy(c->s1);
+  hipStreamDestroy(c->s2);
+  hipEventDestroy(c->e_up);
+  hipEventDestroy(c->e_b1);
+  hipEventDestroy(c->e_b2);
+  delete c;
+}

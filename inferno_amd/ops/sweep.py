@@ -83,6 +83,14 @@ def load_library(allow_build: bool = True) -> ctypes.CDLL:
     lib.wva_sweep_launch_bucket.restype = ctypes.c_int
     lib.wva_argmin_launch.restype = ctypes.c_int
     lib.wva_device_count.restype = ctypes.c_int
+    lib.wva_ctx_create.restype = ctypes.c_void_p
+    lib.wva_ctx_create.argtypes = [ctypes.c_int, ctypes.c_int,
+                                   ctypes.POINTER(ctypes.c_void_p)]
+    lib.wva_ctx_set_buckets.restype = ctypes.c_int
+    lib.wva_reconcile.restype = ctypes.c_int
+    lib.wva_reconcile.argtypes = [ctypes.c_void_p]
+    lib.wva_ctx_destroy.restype = None
+    lib.wva_ctx_destroy.argtypes = [ctypes.c_void_p]
     _lib = lib
     return lib
 
