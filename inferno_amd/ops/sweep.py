@@ -87,6 +87,16 @@ def load_library(allow_build: bool = True) -> ctypes.CDLL:
     lib.wva_ctx_create.argtypes = [ctypes.c_int, ctypes.c_int,
                                    ctypes.POINTER(ctypes.c_void_p)]
     lib.wva_ctx_set_buckets.restype = ctypes.c_int
+    lib.wva_ctx_set_buckets.argtypes = [
+        ctypes.c_void_p,
+        ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int),
+        ctypes.POINTER(ctypes.c_void_p),
+        ctypes.POINTER(ctypes.c_int),
+        ctypes.POINTER(ctypes.c_int),
+        ctypes.c_int,
+        ctypes.c_float,
+    ]
     lib.wva_reconcile.restype = ctypes.c_int
     lib.wva_reconcile.argtypes = [ctypes.c_void_p]
     lib.wva_ctx_destroy.restype = None
