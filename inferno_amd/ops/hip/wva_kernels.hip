@@ -808,6 +808,10 @@ extern "C" int wva_sweep_launch_bucket(
       hipLaunchKernelGGL(wva_sweep_t<256>, dim3(n_blocks), dim3(256), lds, (hipStream_t)stream,
                          in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
       break;
+    case 512:
+      hipLaunchKernelGGL(wva_sweep_t<512>, dim3(n_blocks), dim3(512), lds, (hipStream_t)stream,
+                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
+      break;
     case 1024:
       hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(n_blocks), dim3(1024), lds,
                          (hipStream_t)stream, in, out, n_blocks, cell_ids, max_n,
@@ -1017,6 +1021,10 @@ static int wva_launch_bucket_on(WvaCtx *c, const WvaBucket &b, hipStream_t s) {
       break;
     case 256:
       hipLaunchKernelGGL(wva_sweep_t<256>, dim3(b.n_blocks), dim3(256), lds, s, c->in, c->out,
+                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
+      break;
+    case 512:
+      hipLaunchKernelGGL(wva_sweep_t<512>, dim3(b.n_blocks), dim3(512), lds, s, c->in, c->out,
                          b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
       break;
     case 1024:

@@ -38,23 +38,24 @@ def choose_buckets(batch_n):
 
     n = len(batch_n)
     masks = [
-        (64, batch_n <= N_SMALL, None),
-        (256, (batch_n > N_SMALL) & (batch_n <= N_MED), 64),
-        (1024, batch_n > N_MED, 256),
+        (64, batch_n <= N_SMALL),
+        (256, (batch_n > N_SMALL) & (batch_n <= N_MED)),
+        (1024, batch_n > N_MED),
     ]
     out = []
-    for base_nt, mask, demoted in masks:
+    for base_nt, mask in masks:
         idx = np.nonzero(mask)[0]
         count = len(idx)
         if count == 0:
             continue
         nt = base_nt
-        # demote to the narrower block size when the bucket is
-        # throughput-bound (more blocks resident than the wide shape allows)
-        if demoted is not None:
-            resident = {256: 1024, 1024: 512}[base_nt]
-            if count > resident:
-                nt = demoted
+        # shrink the block until the whole bucket is resident in one
+        # dispatch wave (residency per shape on MI355X: 1024-thread blocks
+        # run 1/CU = 256, 512-thread 2/CU = 512, 256-thread 4/CU = 1024)
+        if base_nt == 1024:
+            nt = 1024 if count <= 256 else (512 if count <= 512 else 256)
+        elif base_nt == 256:
+            nt = 256 if count <= 1024 else 64
         ids = None if count == n else idx.astype(np.int32)
         out.append((nt, ids, int(batch_n[mask].max()), count))
     return out
