@@ -261,7 +261,6 @@ class FastSweep:
         st["pin_out_i"] = torch.empty((4, n_srv), dtype=torch.int32, pin_memory=True)
         st["bucket_key"] = None
         st["buckets"] = []
-        st["side_streams"] = []
         self._gpu = st
 
     def _buckets_for(self, batch_n: np.ndarray):
@@ -280,8 +279,6 @@ class FastSweep:
         ]
         st["bucket_key"] = key
         st["buckets"] = buckets
-        while len(st["side_streams"]) < max(len(buckets) - 1, 0):
-            st["side_streams"].append(torch.cuda.Stream())
         return buckets
 
     def _ensure_ctx(self) -> None:
