@@ -16,7 +16,7 @@ from .build import LIB_PATH, build, needs_build
 
 MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
 # N-bucket thresholds (must match WVA_N_SMALL / WVA_N_MED): cells are
-# dispatched to 64/256/1024-thread blocks by batch size — the straggler
+# dispatched to 64/256/512/1024-thread blocks by batch size and regime — the
 # (large-N) cells' latency sets the sweep wall time, so they get wide blocks,
 # while small cells run one barrier-free wave each.
 N_SMALL = 512

@@ -10,10 +10,14 @@ import argparse
 
 import yaml
 
-from . import AMD_GPUS, GRANITE_13B, LLAMA_8B, LLAMA_70B, MIXTRAL_8X7B, LlmSpec, derive_profile, tp_variant_name
+from . import (AMD_GPUS, DEEPSEEK_V3, GRANITE_13B, LLAMA_8B, LLAMA_70B, LLAMA_405B,
+               LLAMA_405B_FP8, MIXTRAL_8X7B, QWEN_72B, LlmSpec, derive_profile,
+               tp_variant_name)
 
 KNOWN_MODELS = {
-    m.name: m for m in (LLAMA_8B, LLAMA_70B, GRANITE_13B, MIXTRAL_8X7B)
+    m.name: m
+    for m in (LLAMA_8B, LLAMA_70B, LLAMA_405B, LLAMA_405B_FP8, GRANITE_13B,
+              MIXTRAL_8X7B, QWEN_72B, DEEPSEEK_V3)
 }
 
 

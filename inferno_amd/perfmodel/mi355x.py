@@ -104,23 +104,43 @@ AMD_GPUS = {"MI355X": MI355X, "MI325X": MI325X, "MI300X": MI300X}
 
 @dataclass
 class LlmSpec:
-    """Transformer shape for the perf model."""
+    """Transformer shape for the perf model.
+
+    ``active_params_b`` covers MoE models: decode weight streaming reads only
+    the routed experts' parameters per token (plus shared weights), while
+    memory capacity must hold ``params_b`` (all experts) — so alpha/beta/delta
+    derive from the ACTIVE parameter count and KV/memory sizing from the
+    TOTAL. Dense models leave it None (= params_b). ``dtype_bytes`` 2.0 =
+    bf16; 1.0 = OCP fp8 (gfx950's e4m3/e5m2 — weights AND the MFMA rate
+    double, hence the fp8_pflops column of GpuSpec).
+    """
 
     name: str
-    params_b: float  # parameters, billions
+    params_b: float  # TOTAL parameters, billions
     layers: int
     hidden: int
     heads: int
     kv_heads: int
-    dtype_bytes: float = 2.0  # bf16 weights/KV
+    dtype_bytes: float = 2.0  # bf16 weights/KV; 1.0 = fp8
+    active_params_b: float | None = None  # MoE: routed-active params
 
     @property
     def head_dim(self) -> int:
         return self.hidden // self.heads
 
     @property
+    def active_b(self) -> float:
+        return self.active_params_b if self.active_params_b is not None else self.params_b
+
+    @property
     def param_bytes(self) -> float:
+        """TOTAL weight bytes (memory capacity)."""
         return self.params_b * 1e9 * self.dtype_bytes
+
+    @property
+    def active_param_bytes(self) -> float:
+        """ACTIVE weight bytes per token (decode streaming)."""
+        return self.active_b * 1e9 * self.dtype_bytes
 
     def kv_bytes_per_token(self, tp: int) -> float:
         """KV cache bytes per token per TP shard: 2 (K and V) x layers x
@@ -131,8 +151,15 @@ class LlmSpec:
 # reference model shapes (public architecture parameters)
 LLAMA_8B = LlmSpec("llama-3.1-8b", 8.0, 32, 4096, 32, 8)
 LLAMA_70B = LlmSpec("llama-3.1-70b", 70.0, 80, 8192, 64, 8)
+LLAMA_405B = LlmSpec("llama-3.1-405b", 405.0, 126, 16384, 128, 8)
+LLAMA_405B_FP8 = LlmSpec("llama-3.1-405b-fp8", 405.0, 126, 16384, 128, 8,
+                         dtype_bytes=1.0)
 GRANITE_13B = LlmSpec("granite-13b", 13.0, 40, 5120, 40, 40)
-MIXTRAL_8X7B = LlmSpec("mixtral-8x7b", 46.7, 32, 4096, 32, 8)
+MIXTRAL_8X7B = LlmSpec("mixtral-8x7b", 46.7, 32, 4096, 32, 8,
+                       active_params_b=12.9)
+QWEN_72B = LlmSpec("qwen2.5-72b", 72.7, 80, 8192, 64, 8)
+DEEPSEEK_V3 = LlmSpec("deepseek-v3", 671.0, 61, 7168, 128, 128,
+                      dtype_bytes=1.0, active_params_b=37.0)
 
 
 def allreduce_ms(message_bytes: float, tp: int, gpu: GpuSpec) -> float:
@@ -178,18 +205,20 @@ def derive_profile(
     exceeds device memory).
     """
     hbm_bps = gpu.hbm_peak_tbs * 1e12 * gpu.hbm_eff
-    mfma_flops = gpu.bf16_pflops * 1e15 * gpu.mfma_eff
+    pflops = gpu.fp8_pflops if model.dtype_bytes <= 1.0 else gpu.bf16_pflops
+    mfma_flops = pflops * 1e15 * gpu.mfma_eff
 
     # ---- decode ITL = alpha + beta*b ---------------------------------
-    # alpha: stream the weight shard once per token + per-layer all-reduce
-    weight_ms = model.param_bytes / tp / hbm_bps * 1e3
+    # alpha: stream the ACTIVE weight shard once per token + per-layer
+    # all-reduce (MoE models stream only routed experts)
+    weight_ms = model.active_param_bytes / tp / hbm_bps * 1e3
     # decode all-reduce per layer: message = hidden * dtype (batch~1 row)
     ar_decode = model.layers * allreduce_ms(model.hidden * model.dtype_bytes, tp, gpu)
     alpha = weight_ms + ar_decode + sched_overhead_ms * 0.25
     # beta: per extra in-flight request, read its KV (at_tokens context) and
     # do its GEMV compute share
     kv_read_ms = model.kv_bytes_per_token(tp) * at_tokens / hbm_bps * 1e3
-    compute_ms = 2.0 * model.params_b * 1e9 / tp / mfma_flops * 1e3
+    compute_ms = 2.0 * model.active_b * 1e9 / tp / mfma_flops * 1e3
     beta = kv_read_ms + compute_ms
 
     # ---- prefill TTFT = gamma + delta * inTokens * b ------------------
@@ -197,7 +226,7 @@ def derive_profile(
     # Per-token all-reduce bandwidth is negligible vs the GEMM work, so the
     # collective cost appears only as the latency floor in gamma — keeping
     # the model linear exactly like the CR's gamma + delta*tokens*batch form.
-    delta = 2.0 * model.params_b * 1e9 / tp / mfma_flops * 1e3
+    delta = 2.0 * model.active_b * 1e9 / tp / mfma_flops * 1e3
     gamma = sched_overhead_ms + model.layers * allreduce_ms(
         model.hidden * model.dtype_bytes, tp, gpu
     )
