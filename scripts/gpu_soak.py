@@ -44,8 +44,12 @@ def main() -> None:
             min_num_replicas=[0, 1][f % 2],
             n_accelerators=[1, 2, 3][f % 3],
         )
-        a, opt = System.from_spec(make_spec(**kw))
-        b, _ = System.from_spec(make_spec(**kw))
+        analyzer = ["mm1k", "mm1k", "mg1"][f % 3]
+        sa, sb = make_spec(**kw), make_spec(**kw)
+        sa.optimizer.analyzer = analyzer
+        sb.optimizer.analyzer = analyzer
+        a, opt = System.from_spec(sa)
+        b, _ = System.from_spec(sb)
         SweepEngine(backend="cpu").sweep(a)
         SweepEngine(backend="gpu").sweep(b)
         for name in a.servers:
