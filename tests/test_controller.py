@@ -425,3 +425,32 @@ class TestScalingCounter:
              "direction": "up", "reason": "slo_optimization"},
         )
         assert val == 1.0
+
+
+class TestProbeServer:
+    def test_probe_endpoints(self):
+        import urllib.request
+
+        from inferno_amd.controller.main import gpu_health_probe, serve_probes
+
+        state = {"ready": False}
+        serve_probes(18473, state)
+        base = "http://127.0.0.1:18473"
+
+        def get(path):
+            try:
+                with urllib.request.urlopen(base + path, timeout=5) as r:
+                    return r.status, r.read()
+            except urllib.error.HTTPError as e:
+                return e.code, b""
+
+        assert get("/healthz")[0] == 200
+        assert get("/readyz")[0] == 503  # not ready yet
+        state["ready"] = True
+        assert get("/readyz")[0] == 200
+        code, body = get("/metrics")
+        assert code == 200
+        assert get("/nope")[0] == 404
+
+        probe = gpu_health_probe()
+        assert "gpu" in probe  # False in this container, True on an MI355X
