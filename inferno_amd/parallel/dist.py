@@ -312,6 +312,70 @@ class ShardedSolver:
         stats = self.engine.sweep(system, server_names=local_names)
         return self._finish_slow(system, local_names, spec, acc_names, stats)
 
+    def _allgather_candidates(self, system: System, local_names, acc_names, world):
+        """Greedy limited mode at world>1: all-gather every rank's candidate
+        allocations so each rank runs the IDENTICAL global greedy (capacity
+        is a global constraint — rank-local greedy would double-spend it).
+        Candidate record: [server_gidx, acc_idx, zero_empty, replicas, batch,
+        cost, value, itl, ttft, rho, max_rate] (f32)."""
+        import torch
+        import torch.distributed as dist
+
+        from ..core import Allocation
+
+        all_names = sorted(system.servers)
+        name_to_global = {n: i for i, n in enumerate(all_names)}
+        acc_index = {n: i for i, n in enumerate(acc_names)}
+        recs = []
+        for name in local_names:
+            gidx = float(name_to_global[name])
+            for acc_key, alloc in system.servers[name].all_allocations.items():
+                recs.append([
+                    gidx,
+                    float(acc_index[acc_key]),
+                    1.0 if alloc.accelerator == "" else 0.0,
+                    float(alloc.num_replicas),
+                    float(alloc.batch_size),
+                    alloc.cost,
+                    alloc.value,
+                    alloc.itl,
+                    alloc.ttft,
+                    alloc.rho,
+                    alloc.max_arrv_rate_per_replica,
+                ])
+        local = torch.tensor(recs, dtype=torch.float32).reshape(-1, 11)
+        backend = dist.get_backend(self.group)
+        comm_dev = "cuda" if backend == "nccl" else "cpu"
+        local = local.to(comm_dev)
+        # variable-size all_gather: exchange lengths first
+        n_local = torch.tensor([local.shape[0]], dtype=torch.int64, device=comm_dev)
+        sizes = [torch.zeros_like(n_local) for _ in range(world)]
+        dist.all_gather(sizes, n_local, group=self.group)
+        max_n = int(max(s.item() for s in sizes))
+        padded = torch.zeros((max_n, 11), dtype=torch.float32, device=comm_dev)
+        padded[: local.shape[0]] = local
+        gathered = [torch.empty_like(padded) for _ in range(world)]
+        dist.all_gather(gathered, padded, group=self.group)
+        # rebuild every server's candidate map (identical on all ranks)
+        for name in all_names:
+            system.servers[name].all_allocations = {}
+        for r, g in enumerate(gathered):
+            rows = g[: int(sizes[r].item())].cpu().numpy()
+            for row in rows:
+                name = all_names[int(row[0])]
+                acc_key = acc_names[int(row[1])]
+                system.servers[name].all_allocations[acc_key] = Allocation(
+                    accelerator="" if row[2] > 0.5 else acc_key,
+                    num_replicas=int(row[3]),
+                    batch_size=int(row[4]),
+                    cost=float(row[5]),
+                    value=float(row[6]),
+                    itl=float(row[7]),
+                    ttft=float(row[8]),
+                    rho=float(row[9]),
+                    max_arrv_rate_per_replica=float(row[10]),
+                )
+
     def _finish_slow(self, system: System, local_names, spec: OptimizerSpec, acc_names,
                      stats):
         if spec.unlimited:
@@ -326,12 +390,14 @@ class ShardedSolver:
                 if best is not None:
                     server.set_allocation(best)
         else:
-            # greedy limited mode needs the full candidate lists; capacity is
-            # global, so greedy over a shard is only exact for world==1 —
-            # multi-rank greedy runs rank-local greedy on the shard with the
-            # full capacity map (documented approximation).
+            # greedy limited mode needs the full candidate lists and capacity
+            # is a GLOBAL constraint: at world>1 each rank all-gathers every
+            # shard's candidates and runs the identical global greedy
             from ..solver.greedy import solve_greedy
 
+            initialized, _rank, world = self._dist_info()
+            if initialized and world > 1:
+                self._allgather_candidates(system, local_names, acc_names, world)
             solve_greedy(
                 system,
                 delayed_best_effort=spec.delayedBestEffort,
