@@ -102,3 +102,33 @@ class TestExampleArtifacts:
         assert vals["controller"]["backend"] in ("auto", "gpu", "cpu")
         chart = load_yaml_docs("charts/workload-variant-autoscaler/Chart.yaml")[0]
         assert chart["name"] == "workload-variant-autoscaler"
+
+
+class TestOfflineCli:
+    def test_solve_and_analyze(self, tmp_path, capsys):
+        import json as _json
+        import sys as _sys
+
+        from inferno_amd.config import system_spec_to_json
+        from inferno_amd import cli
+        from tests.fixtures import make_spec
+
+        path = tmp_path / "sys.json"
+        path.write_text(_json.dumps(system_spec_to_json(make_spec(n_servers=3, seed=56))))
+
+        class A:  # solve args
+            spec = str(path)
+            backend = "cpu"
+            json = True
+
+        assert cli.cmd_solve(A()) == 0
+        out = _json.loads(capsys.readouterr().out)
+        assert len(out["allocations"]) == 3
+
+        class B:  # analyze args
+            spec = str(path)
+            backend = "cpu"
+            server = "srv-1:ns"
+
+        assert cli.cmd_analyze(B()) == 0
+        assert "accelerator" in capsys.readouterr().out
