@@ -100,7 +100,13 @@ class Configuration:
     service_parms: ServiceParms
 
     def check(self) -> None:
-        if self.max_batch_size <= 0 or self.max_queue_size < 0 or self.service_parms is None:
+        if (
+            self.max_batch_size <= 0
+            or self.max_queue_size < 0
+            or self.service_parms is None
+            or self.service_parms.prefill is None
+            or self.service_parms.decode is None
+        ):
             raise AnalyzerError(f"invalid configuration {self}")
 
 
