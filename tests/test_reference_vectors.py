@@ -297,3 +297,174 @@ class TestBinarySearch:
         assert ind == 0
         assert abs(x - 3.0) <= 0.01
         assert abs(x * x - 9.0) <= 0.1
+
+
+# ---------------------------------------------------------------------------
+# Core-layer vectors (pkg/core/accelerator_test.go, pkg/core/allocation_test.go)
+# ---------------------------------------------------------------------------
+
+from types import SimpleNamespace
+
+from inferno_amd.config import (
+    AcceleratorSpec,
+    DecodeParms as CfgDecodeParms,
+    ModelAcceleratorPerfData,
+    PowerSpec,
+    PrefillParms as CfgPrefillParms,
+)
+from inferno_amd.core.allocation import Allocation, _zero_load_allocation
+from inferno_amd.core.system import Accelerator, Model
+
+
+class TestAcceleratorPower:
+    """Ref: accelerator_test.go TestAccelerator_Power — piecewise-linear."""
+
+    @pytest.mark.parametrize(
+        "util,want",
+        [
+            (0.0, 100.0),   # idle power
+            (0.5, 300.0),   # mid power
+            (1.0, 700.0),   # full power
+            (0.25, 200.0),  # interpolated idle..mid
+            (0.75, 500.0),  # interpolated mid..full
+        ],
+    )
+    def test_vectors(self, util, want):
+        acc = Accelerator(AcceleratorSpec(
+            name="TestAcc",
+            power=PowerSpec(idle=100, midPower=300, full=700, midUtil=0.5),
+        ))
+        acc.calculate()
+        assert acc.power(util) == pytest.approx(want)
+
+
+class TestTransitionPenalty:
+    """Ref: allocation_test.go TestAllocation_TransitionPenalty."""
+
+    def test_same_accelerator_same_replicas(self):
+        a = Allocation(accelerator="gpu-a", num_replicas=2, cost=100.0)
+        b = Allocation(accelerator="gpu-a", num_replicas=2, cost=100.0)
+        assert a.transition_penalty(b) == 0.0
+
+    def test_same_accelerator_different_replicas(self):
+        a = Allocation(accelerator="gpu-a", num_replicas=2, cost=100.0)
+        b = Allocation(accelerator="gpu-a", num_replicas=3, cost=150.0)
+        assert a.transition_penalty(b) == pytest.approx(50.0)  # cost difference
+
+    def test_different_accelerator(self):
+        a = Allocation(accelerator="gpu-a", num_replicas=2, cost=100.0)
+        b = Allocation(accelerator="gpu-b", num_replicas=2, cost=120.0)
+        # 0.1*(100+120) + (120-100)
+        assert a.transition_penalty(b) == pytest.approx(0.1 * 220.0 + 20.0)
+
+
+class TestZeroLoadAllocation:
+    """Ref: allocation_test.go TestZeroLoadAllocation."""
+
+    @staticmethod
+    def _model(name, instances):
+        m = Model(name)
+        m.num_instances = dict(instances)
+        return m
+
+    def test_zero_replicas_is_empty(self):
+        server = SimpleNamespace(min_num_replicas=0, max_batch_size=0)
+        model = self._model("test-model", {})
+        acc = Accelerator(AcceleratorSpec(name="test-gpu", cost=100.0))
+        perf = ModelAcceleratorPerfData(
+            maxBatchSize=16,
+            decodeParms=CfgDecodeParms(alpha=5.0, beta=2.0),
+            prefillParms=CfgPrefillParms(gamma=10.0, delta=1.5),
+        )
+        alloc = _zero_load_allocation(server, model, acc, perf)
+        assert alloc.accelerator == ""
+        assert alloc.num_replicas == 0
+        assert alloc.batch_size == 0
+        assert alloc.cost == 0.0
+        assert alloc.value == alloc.cost
+        assert alloc.rho == 0
+
+    def test_min_replicas(self):
+        server = SimpleNamespace(min_num_replicas=2, max_batch_size=0)
+        model = self._model("test-model", {"test-gpu": 1})
+        acc = Accelerator(AcceleratorSpec(name="test-gpu", cost=100.0))
+        perf = ModelAcceleratorPerfData(
+            maxBatchSize=16,
+            decodeParms=CfgDecodeParms(alpha=5.0, beta=2.0),
+            prefillParms=CfgPrefillParms(gamma=10.0, delta=1.5),
+        )
+        alloc = _zero_load_allocation(server, model, acc, perf)
+        assert alloc.accelerator == "test-gpu"
+        assert alloc.num_replicas == 2
+        assert alloc.batch_size == 16
+        assert alloc.cost == pytest.approx(200.0)  # 100 * 1 instance * 2 replicas
+        assert alloc.value == alloc.cost
+        assert alloc.rho == 0
+        assert alloc.itl == pytest.approx(5.0 + 2.0)      # alpha + beta
+        assert alloc.ttft == pytest.approx(10.0 + 1.5)    # gamma + delta
+        max_decode = 5.0 + 2.0 * 16
+        max_serv = (10.0 + 1.5) + max_decode
+        assert alloc.max_arrv_rate_per_replica == pytest.approx(16.0 / max_serv)
+
+    def test_server_batch_override(self):
+        server = SimpleNamespace(min_num_replicas=1, max_batch_size=8)
+        model = self._model("test-model", {"test-gpu": 2})
+        acc = Accelerator(AcceleratorSpec(name="test-gpu", cost=50.0))
+        perf = ModelAcceleratorPerfData(
+            maxBatchSize=16,  # overridden by server.max_batch_size
+            decodeParms=CfgDecodeParms(alpha=3.0, beta=1.0),
+            prefillParms=CfgPrefillParms(gamma=8.0, delta=2.0),
+        )
+        alloc = _zero_load_allocation(server, model, acc, perf)
+        assert alloc.batch_size == 8
+        assert alloc.num_replicas == 1
+        assert alloc.cost == pytest.approx(100.0)  # 50 * 2 instances * 1 replica
+
+    def test_minimal_inputs_do_not_crash(self):
+        server = SimpleNamespace(min_num_replicas=1, max_batch_size=0)
+        model = self._model("m", {})
+        acc = Accelerator(AcceleratorSpec(name="test-gpu", cost=0.0))
+        perf = ModelAcceleratorPerfData(
+            maxBatchSize=1,
+            decodeParms=CfgDecodeParms(alpha=0.1, beta=0.1),
+            prefillParms=CfgPrefillParms(gamma=0.1, delta=0.1),
+        )
+        assert _zero_load_allocation(server, model, acc, perf) is not None
+
+
+class TestServerPriority:
+    """Ref: server_test.go TestServer_Priority + serviceclass clamping."""
+
+    def _system(self):
+        from inferno_amd.core.system import ServiceClass, System
+
+        system = System()
+        system.service_classes["high-priority"] = ServiceClass("high-priority", 1)
+        system.service_classes["low-priority"] = ServiceClass("low-priority", 8)
+        return system
+
+    @pytest.mark.parametrize(
+        "klass,expected",
+        [
+            ("high-priority", 1),
+            ("low-priority", 8),
+            ("nonexistent", 100),  # DefaultServiceClassPriority = DefaultLowPriority
+        ],
+    )
+    def test_priority(self, klass, expected):
+        from inferno_amd.config import AllocationData, ServerSpec
+        from inferno_amd.core.system import Server
+
+        server = Server(ServerSpec(
+            name="test-server", model="test-model", klass=klass,
+            currentAlloc=AllocationData(),
+        ))
+        assert server.priority(self._system()) == expected
+
+    def test_out_of_range_priority_clamps_to_default(self):
+        from inferno_amd.core.system import ServiceClass
+
+        # priority outside [DefaultHighPriority=1, DefaultLowPriority=100]
+        assert ServiceClass("weird", 0).priority == 100
+        assert ServiceClass("weird", 101).priority == 100
+        assert ServiceClass("ok", 50).priority == 50
