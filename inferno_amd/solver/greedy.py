@@ -86,9 +86,15 @@ def _allocate(
     Ref: greedy.go:107-166.
     """
     entries = list(entries)
+    # parallel sorted key list so re-insertion is a bisect over cached keys
+    # instead of re-materializing every key on each pass (the reference's
+    # slices.BinarySearchFunc recomputes orderFunc per comparison, greedy.go
+    # :158-162 — same ordering, cheaper bookkeeping)
+    keys = [_order_key(e) for e in entries]
     unallocated: list[ServerEntry] = []
     while entries:
         top = entries.pop(0)
+        keys.pop(0)
         if not top.allocations:
             continue
         server = system.servers.get(top.server_name)
@@ -120,9 +126,10 @@ def _allocate(
                 continue
             else:
                 top.delta = _MAX_FLOAT32
-            keys = [_order_key(e) for e in entries]
-            i = bisect.bisect_left(keys, _order_key(top))
+            k = _order_key(top)
+            i = bisect.bisect_left(keys, k)
             entries.insert(i, top)
+            keys.insert(i, k)
     return unallocated
 
 
