@@ -268,16 +268,23 @@ class StateDependentChain:
         wN = float(w[N])
         Q = K - N  # number of tail states
         if Q > 0 and wN > 0.0:
-            if abs(1.0 - r) < 1e-12:
-                # r == 1: flat tail
+            rQ = math.exp(Q * log_r)
+            if Q * abs(log_r) < 1e-6:
+                # r ~ 1: flat tail (relative error < Q|log r|/2 < 5e-7); the
+                # closed forms below cancel catastrophically in this regime
                 tail_sum = wN * Q
                 tail_n_sum = wN * (Q * N + Q * (Q + 1) / 2.0)
-                wK = wN
+                wK = wN * rQ
             else:
-                rQ = math.exp(Q * log_r)
-                g = r * (1.0 - rQ) / (1.0 - r)  # sum_{j=1..Q} r^j
+                # expm1-stable forms: 1-r and 1-r^Q computed without
+                # cancellation, and the arithmetico-geometric numerator
+                # rewritten as (1-r^Q) - Q r^Q (1-r) so its error is
+                # O(eps / (Q|log r|)) <= ~4e-10 at the 1e-6 threshold
+                one_m_r = -math.expm1(log_r)
+                one_m_rQ = -math.expm1(Q * log_r)
+                g = r * one_m_rQ / one_m_r  # sum_{j=1..Q} r^j
                 # sum_{j=1..Q} j r^j (arithmetico-geometric)
-                jg = r * (1.0 - (Q + 1) * rQ + Q * rQ * r) / ((1.0 - r) ** 2)
+                jg = r * (one_m_rQ - Q * rQ * one_m_r) / (one_m_r * one_m_r)
                 tail_sum = wN * g
                 tail_n_sum = wN * (N * g + jg)
                 wK = wN * rQ

@@ -287,15 +287,19 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
   const int Q = K - N;
   double tail_sum = 0.0, tail_n_sum = 0.0, wK = (Q == 0) ? wN : 0.0;
   if (Q > 0 && wN > 0.0) {
-    if (fabs(1.0 - r) < 1e-12) {
+    const double rQ = exp((double)Q * log_r);
+    if ((double)Q * fabs(log_r) < 1e-6) {
+      // r ~ 1: flat tail; the closed forms cancel catastrophically here
       tail_sum = wN * (double)Q;
       tail_n_sum = wN * ((double)Q * (double)N + (double)Q * (double)(Q + 1) * 0.5);
-      wK = wN;
+      wK = wN * rQ;
     } else {
-      const double rQ = exp((double)Q * log_r);
-      const double omr = 1.0 - r;
-      const double gg = r * (1.0 - rQ) / omr;
-      const double jg = r * (1.0 - (double)(Q + 1) * rQ + (double)Q * rQ * r) / (omr * omr);
+      // expm1-stable: 1-r, 1-r^Q without cancellation; arithmetico-geometric
+      // numerator rewritten as (1-r^Q) - Q r^Q (1-r)
+      const double omr = -expm1(log_r);
+      const double omrQ = -expm1((double)Q * log_r);
+      const double gg = r * omrQ / omr;
+      const double jg = r * (omrQ - (double)Q * rQ * omr) / (omr * omr);
       tail_sum = wN * gg;
       tail_n_sum = wN * ((double)N * gg + jg);
       wK = wN * rQ;
@@ -1085,11 +1089,11 @@ extern "C" int wva_reconcile(void *ctx) {
 extern "C" void wva_ctx_destroy(void *ctx) {
   WvaCtx *c = (WvaCtx *)ctx;
   if (c == nullptr) return;
-  hipStreamDestroy(c->s0);
-  hipStreamDestroy(c->s1);
-  hipStreamDestroy(c->s2);
-  hipEventDestroy(c->e_up);
-  hipEventDestroy(c->e_b1);
-  hipEventDestroy(c->e_b2);
+  (void)hipStreamDestroy(c->s0);
+  (void)hipStreamDestroy(c->s1);
+  (void)hipStreamDestroy(c->s2);
+  (void)hipEventDestroy(c->e_up);
+  (void)hipEventDestroy(c->e_b1);
+  (void)hipEventDestroy(c->e_b2);
   delete c;
 }

@@ -294,3 +294,74 @@ class TestMM1K:
         a = MM1K(10).solve(0.5, 1.0)
         b = MG1K(10, cv2=0.0).solve(0.5, 1.0)
         assert b.avg_wait_time == pytest.approx(0.5 * a.avg_wait_time, rel=1e-12)
+
+
+class TestFlatTailBranch:
+    """The r = lam/s(N) == 1 geometric-tail special case (flat tail): the
+    closed form switches to the arithmetic branch; it must agree with the
+    reference-style forward recurrence exactly like the generic branch."""
+
+    def test_flat_tail_matches_oracle(self):
+        import numpy as np
+
+        from inferno_amd.analyzer.queue import StateDependentChain
+        from tests.oracle import chain_stats_recurrence
+
+        serv = np.linspace(0.05, 0.4, 16).astype(np.float32)
+        K = 11 * len(serv)
+        chain = StateDependentChain(K, serv)
+        lam = float(serv[-1])  # exactly s(N): r == 1 -> flat tail
+        got = chain.solve(lam)
+        want = chain_stats_recurrence(K, serv, lam)
+        assert got.throughput == pytest.approx(want["throughput"], rel=1e-9)
+        assert got.avg_num_in_system == pytest.approx(
+            want["avg_num_in_system"], rel=1e-9)
+        assert got.pK == pytest.approx(want["pK"], rel=1e-9)
+        assert got.p0 == pytest.approx(want["p0"], rel=1e-9)
+
+    def test_near_flat_tail_continuity(self):
+        """Values just below/above r=1 bracket the flat-tail value (the
+        branch switch introduces no discontinuity)."""
+        import numpy as np
+
+        from inferno_amd.analyzer.queue import StateDependentChain
+
+        serv = np.linspace(0.05, 0.4, 16).astype(np.float32)
+        chain = StateDependentChain(11 * len(serv), serv)
+        sN = float(serv[-1])
+        at = chain.solve(sN).avg_num_in_system
+        below = chain.solve(sN * (1 - 1e-9)).avg_num_in_system
+        above = chain.solve(sN * (1 + 1e-9)).avg_num_in_system
+        assert below <= at <= above or above <= at <= below
+        assert at == pytest.approx(below, rel=1e-5)
+        assert at == pytest.approx(above, rel=1e-5)
+
+    def test_zero_queue_K_equals_N(self):
+        """K == N (no queue states): tail sums vanish; compare to oracle."""
+        import numpy as np
+
+        from inferno_amd.analyzer.queue import StateDependentChain
+        from tests.oracle import chain_stats_recurrence
+
+        serv = np.linspace(0.1, 0.8, 8).astype(np.float32)
+        chain = StateDependentChain(len(serv), serv)
+        lam = 0.3
+        got = chain.solve(lam)
+        want = chain_stats_recurrence(len(serv), serv, lam)
+        assert got.throughput == pytest.approx(want["throughput"], rel=1e-9)
+        assert got.pK == pytest.approx(want["pK"], rel=1e-9)
+
+    def test_single_state_chain(self):
+        """N=1, K=11: pure M/M/1/K degenerate chain vs oracle."""
+        import numpy as np
+
+        from inferno_amd.analyzer.queue import StateDependentChain
+        from tests.oracle import chain_stats_recurrence
+
+        serv = np.array([0.25], dtype=np.float32)
+        chain = StateDependentChain(11, serv)
+        for lam in (0.01, 0.2, 0.249):
+            got = chain.solve(lam)
+            want = chain_stats_recurrence(11, serv, lam)
+            assert got.avg_num_in_system == pytest.approx(
+                want["avg_num_in_system"], rel=1e-9)
