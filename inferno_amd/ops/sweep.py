@@ -56,6 +56,12 @@ def choose_buckets(batch_n):
             nt = 1024 if count <= 256 else (512 if count <= 512 else 256)
         elif base_nt == 256:
             nt = 256 if count <= 1024 else 64
+        # measurement override (A/B tuning): INFERNO_NT_SMALL/_MED/_LARGE
+        env = os.environ.get(
+            {64: "INFERNO_NT_SMALL", 256: "INFERNO_NT_MED", 1024: "INFERNO_NT_LARGE"}[base_nt]
+        )
+        if env:
+            nt = int(env)
         ids = None if count == n else idx.astype(np.int32)
         out.append((nt, ids, int(batch_n[mask].max()), count))
     return out
