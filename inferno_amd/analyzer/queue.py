@@ -196,6 +196,11 @@ def binary_search(x_min, x_max, y_target, eval_fn):
         y_star = eval_fn(x_star)
         if within_tolerance(y_star, y_target, SEARCH_TOLERANCE):
             break
+        # fixed point: interval collapsed to <=1 ulp — every remaining
+        # iteration re-evaluates this same midpoint until the cap, so the
+        # result is already final (identical x_star, fewer evaluations)
+        if x_star == x_min or x_star == x_max:
+            break
         if (increasing and y_target < y_star) or (not increasing and y_target > y_star):
             x_max = x_star
         else:

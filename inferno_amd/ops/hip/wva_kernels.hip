@@ -454,7 +454,12 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
     } else if (phase == 2) {
       x_star = x_eval;
       ++iters;
-      if (within_tol(y, target) || iters >= WVA_MAX_ITERS) {
+      // fixed point: the interval has collapsed to <=1 ulp, so the midpoint
+      // equals an endpoint and EVERY remaining iteration re-evaluates this
+      // same x with the same outcome until the cap — exiting now returns the
+      // identical x_star/ind with up to ~45 fewer chain evaluations
+      const bool fixed_pt = (x_eval == x_min) || (x_eval == x_max);
+      if (within_tol(y, target) || iters >= WVA_MAX_ITERS || fixed_pt) {
         phase = 3;
       } else if ((increasing && target < y) || (!increasing && target > y)) {
         x_max = x_star;

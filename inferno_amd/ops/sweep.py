@@ -48,14 +48,16 @@ def choose_buckets(batch_n):
         count = len(idx)
         if count == 0:
             continue
-        nt = base_nt
-        # shrink the block until the whole bucket is resident in one
-        # dispatch wave (residency per shape on MI355X: 1024-thread blocks
-        # run 1/CU = 256, 512-thread 2/CU = 512, 256-thread 4/CU = 1024)
-        if base_nt == 1024:
-            nt = 1024 if count <= 256 else (512 if count <= 512 else 256)
-        elif base_nt == 256:
-            nt = 256 if count <= 1024 else 64
+        # Measured on MI355X (profiles/REPORT.md, 512-model fleet A/B): the
+        # per-eval cost of a bisection step is dominated by the REDUNDANT
+        # per-lane work (log/exp/expm1 of the tail, loop headers) which costs
+        # waves-per-SIMD x instructions — 16-wave blocks pay 4x what 4-wave
+        # blocks pay — while the serial per-lane chain states are cheap fp32
+        # LDS FMAs. So each bucket wants the NARROWEST block whose serial
+        # states/lane stay modest (~64): N<=512 -> 64 threads, N<=2048 -> 64
+        # (66 states/lane), N<=8192 -> 256 (66 states/lane). 1024-thread
+        # blocks lost 25-43% end-to-end in the A/B (ab_large_*/ab2_* runs).
+        nt = {64: 64, 256: 64, 1024: 256}[base_nt]
         # measurement override (A/B tuning): INFERNO_NT_SMALL/_MED/_LARGE
         env = os.environ.get(
             {64: "INFERNO_NT_SMALL", 256: "INFERNO_NT_MED", 1024: "INFERNO_NT_LARGE"}[base_nt]
