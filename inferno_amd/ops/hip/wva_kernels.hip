@@ -232,6 +232,50 @@ struct ChainGeom {
   int ksub;                // anchors per lane = ceil(chunk/SUB)
 };
 
+// Geometric tail (n = N+1..K, ratio r = lam/s(N)) + final chain statistics
+// from the reduced head sums. Pure scalar fp64; factored out so chain_eval
+// can run it on one wave per part and broadcast (see epilogue there).
+__device__ __forceinline__ ChainOut chain_tail_stats(double lam, double loglam, double logsN,
+                                                     double S_total, int N, int K, double m,
+                                                     double head_sum, double head_n_sum) {
+  const double log_r = loglam - logsN;
+  const double r = exp(log_r);
+  const double wN = exp((double)N * loglam - S_total - m);
+  const int Q = K - N;
+  double tail_sum = 0.0, tail_n_sum = 0.0, wK = (Q == 0) ? wN : 0.0;
+  if (Q > 0 && wN > 0.0) {
+    const double rQ = exp((double)Q * log_r);
+    if ((double)Q * fabs(log_r) < 1e-6) {
+      // r ~ 1: flat tail; the closed forms cancel catastrophically here
+      tail_sum = wN * (double)Q;
+      tail_n_sum = wN * ((double)Q * (double)N + (double)Q * (double)(Q + 1) * 0.5);
+      wK = wN * rQ;
+    } else {
+      // expm1-stable: 1-r, 1-r^Q without cancellation; arithmetico-geometric
+      // numerator rewritten as (1-r^Q) - Q r^Q (1-r)
+      const double omr = -expm1(log_r);
+      const double omrQ = -expm1((double)Q * log_r);
+      const double gg = r * omrQ / omr;
+      const double jg = r * (omrQ - (double)Q * rQ * omr) / (omr * omr);
+      tail_sum = wN * gg;
+      tail_n_sum = wN * ((double)N * gg + jg);
+      wK = wN * rQ;
+    }
+  }
+  const double Z = head_sum + tail_sum;
+  const double pK = wK / Z;
+  const double avg_n_sys = (head_n_sum + tail_n_sum) / Z;
+  const double avg_n_serv = head_n_sum / Z + (1.0 - head_sum / Z) * (double)N;
+
+  ChainOut o;
+  o.throughput = lam * (1.0 - pK);
+  o.in_servers = avg_n_serv;
+  const double resp = avg_n_sys / o.throughput;
+  o.serv = avg_n_serv / o.throughput;
+  o.wait = fmax(resp - o.serv, 0.0);
+  return o;
+}
+
 // Solve the state-dependent chain at arrival rate lam. With PART=1 the whole
 // block cooperates and every thread returns identical results; with PART=2
 // each half-block independently evaluates its own lam (each half covers all
@@ -280,42 +324,36 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
   }
   red_sum2_p<NT, PART>(head_sum, head_n_sum, scratch);
 
-  // geometric tail n = N+1..K with ratio r = lam/s(N)
-  const double log_r = loglam - logsN;
-  const double r = exp(log_r);
-  const double wN = exp((double)N * loglam - g.S_total - m);
-  const int Q = K - N;
-  double tail_sum = 0.0, tail_n_sum = 0.0, wK = (Q == 0) ? wN : 0.0;
-  if (Q > 0 && wN > 0.0) {
-    const double rQ = exp((double)Q * log_r);
-    if ((double)Q * fabs(log_r) < 1e-6) {
-      // r ~ 1: flat tail; the closed forms cancel catastrophically here
-      tail_sum = wN * (double)Q;
-      tail_n_sum = wN * ((double)Q * (double)N + (double)Q * (double)(Q + 1) * 0.5);
-      wK = wN * rQ;
-    } else {
-      // expm1-stable: 1-r, 1-r^Q without cancellation; arithmetico-geometric
-      // numerator rewritten as (1-r^Q) - Q r^Q (1-r)
-      const double omr = -expm1(log_r);
-      const double omrQ = -expm1((double)Q * log_r);
-      const double gg = r * omrQ / omr;
-      const double jg = r * (omrQ - (double)Q * rQ * omr) / (omr * omr);
-      tail_sum = wN * gg;
-      tail_n_sum = wN * ((double)N * gg + jg);
-      wK = wN * rQ;
-    }
-  }
-  const double Z = head_sum + tail_sum;
-  const double pK = wK / Z;
-  const double avg_n_sys = (head_n_sum + tail_n_sum) / Z;
-  const double avg_n_serv = head_n_sum / Z + (1.0 - head_sum / Z) * (double)N;
-
+  // tail + final statistics: identical on every lane of the part (the head
+  // sums are reduction results), so for multi-wave parts only the FIRST wave
+  // computes it and broadcasts through LDS — the transcendental-heavy block
+  // (log/exp/expm1) otherwise costs waves-per-part x instructions of VALU
+  // throughput (the measured reason narrow blocks beat wide ones; see
+  // ops/sweep.py choose_buckets)
+  constexpr int NWH_T = (NT / PART) / WVA_WAVE;
   ChainOut o;
-  o.throughput = lam * (1.0 - pK);
-  o.in_servers = avg_n_serv;
-  const double resp = avg_n_sys / o.throughput;
-  o.serv = avg_n_serv / o.throughput;
-  o.wait = fmax(resp - o.serv, 0.0);
+  if constexpr (NWH_T > 1) {
+    __syncthreads();  // scratch handoff from red_sum2_p
+    if (lane < WVA_WAVE) {
+      o = chain_tail_stats(lam, loglam, logsN, g.S_total, N, K, m, head_sum, head_n_sum);
+      if (lane == 0) {
+        const int part = threadIdx.x / W;
+        scratch[part * 4 + 0] = o.throughput;
+        scratch[part * 4 + 1] = o.wait;
+        scratch[part * 4 + 2] = o.serv;
+        scratch[part * 4 + 3] = o.in_servers;
+      }
+    }
+    __syncthreads();
+    const int part = threadIdx.x / W;
+    o.throughput = scratch[part * 4 + 0];
+    o.wait = scratch[part * 4 + 1];
+    o.serv = scratch[part * 4 + 2];
+    o.in_servers = scratch[part * 4 + 3];
+  } else {
+    // single-wave part: lockstep execution makes the redundancy free
+    o = chain_tail_stats(lam, loglam, logsN, g.S_total, N, K, m, head_sum, head_n_sum);
+  }
   return o;
 }
 
