@@ -315,7 +315,31 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
       head_sum += w;
       head_n_sum += (double)na * w;
       const int jend = min(j0 + WVA_SUB - 1, n1c - n0c);
-      for (int j = j0 + 1; j <= jend; ++j) {
+      // batch the LDS reads 4-wide ahead of the dependent w-chain: with the
+      // narrow-block dispatch there is often only one wave per SIMD, so an
+      // un-batched loop exposes the full LDS latency on every step (PMC:
+      // SQ_WAIT_ANY was 50% of wave cycles). Arithmetic order is unchanged
+      // (bit-identical results).
+      int j = j0 + 1;
+      for (; j + 3 <= jend; j += 4) {
+        const double f0 = (double)g.inv_s_t[j * NT + c];
+        const double f1 = (double)g.inv_s_t[(j + 1) * NT + c];
+        const double f2 = (double)g.inv_s_t[(j + 2) * NT + c];
+        const double f3 = (double)g.inv_s_t[(j + 3) * NT + c];
+        w *= lam * f0;
+        head_sum += w;
+        head_n_sum += (double)(n0c + j) * w;
+        w *= lam * f1;
+        head_sum += w;
+        head_n_sum += (double)(n0c + j + 1) * w;
+        w *= lam * f2;
+        head_sum += w;
+        head_n_sum += (double)(n0c + j + 2) * w;
+        w *= lam * f3;
+        head_sum += w;
+        head_n_sum += (double)(n0c + j + 3) * w;
+      }
+      for (; j <= jend; ++j) {
         w *= lam * (double)g.inv_s_t[j * NT + c];
         head_sum += w;
         head_n_sum += (double)(n0c + j) * w;
