@@ -7,11 +7,12 @@
 // pkg/solver/solver.go:63-79 SolveUnlimited) with batched device kernels:
 //
 //   K1 wva_sweep_t<NT>: one block per (server, accelerator[, TP]) cell,
-//       templated on block size. The host dispatches cells to 64/256/1024-
-//       thread blocks by batch size N AND by regime (ops/sweep.py
-//       choose_buckets): wide blocks shorten a single cell's latency and are
-//       used only when the bucket is small enough to be latency-bound; in
-//       the throughput regime (cells >> CUs) narrower blocks win residency.
+//       templated on block size (64/256/512/1024 instantiated). The host
+//       dispatches cells to 64- and 256-thread blocks by batch size N
+//       (ops/sweep.py choose_buckets) — widths picked by A/B measurement:
+//       per-evaluation cost is dominated by the redundant per-lane
+//       transcendental tail, which costs waves-per-SIMD x instructions, so
+//       NARROW blocks win (16-wave blocks pay 4x what 4-wave blocks pay).
 //       The buckets launch on separate HIP streams so they overlap.
 //
 //       Per cell (state-dependent M/M/1/K mode, the reference's evaluator):
