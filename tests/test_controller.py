@@ -454,3 +454,81 @@ class TestProbeServer:
 
         probe = gpu_health_probe()
         assert "gpu" in probe  # False in this container, True on an MI355X
+
+
+class TestVaValidationEdges:
+    """Mirrors the reference controller specs (variantautoscaling_controller_
+    test.go:444-534): empty modelID / empty accelerator list / empty
+    sloClassRef are handled gracefully without breaking other VAs."""
+
+    def test_empty_model_id_skipped(self):
+        kube, prom, em, reg, rec = build_world()
+        bad = make_va(name="empty-model", model="")
+        bad.spec.modelID = ""
+        kube.add_va(bad)
+        kube.add_deployment(Deployment(name="empty-model", namespace="default",
+                                       replicas=1, status_replicas=1, uid="uid-em"))
+        result = rec.reconcile()
+        # the good VA still processes; the empty-modelID one is skipped
+        assert result.processed == 1
+        good = kube.vas[("default", "vllme-deploy")]
+        assert api.is_condition_true(good, api.TYPE_OPTIMIZATION_READY)
+
+    def test_empty_accelerator_list_does_not_break_others(self):
+        kube, prom, em, reg, rec = build_world()
+        bad = make_va(name="no-accs", model="default/default")
+        bad.spec.modelProfile = api.ModelProfile(accelerators=[])
+        kube.add_va(bad)
+        kube.add_deployment(Deployment(name="no-accs", namespace="default",
+                                       replicas=1, status_replicas=1, uid="uid-na"))
+        result = rec.reconcile()
+        good = kube.vas[("default", "vllme-deploy")]
+        assert api.is_condition_true(good, api.TYPE_OPTIMIZATION_READY)
+        assert result.processed >= 1
+
+    def test_empty_slo_class_ref_uses_model_lookup(self):
+        """The reconciler resolves SLOs by scanning the service-class
+        ConfigMap for the model (FindModelSLO semantics) — an empty
+        sloClassRef name does not crash the loop."""
+        kube, prom, em, reg, rec = build_world()
+        va = kube.vas[("default", "vllme-deploy")]
+        va.spec.sloClassRef = api.ConfigMapKeyRef(name="", key="")
+        result = rec.reconcile()
+        assert result.processed == 1
+
+
+class TestPrometheusConfigFromEnv:
+    """Mirrors controller specs :253-410 — config resolution precedence and
+    TLS defaults."""
+
+    def test_missing_base_url_empty(self, monkeypatch):
+        for k in ("PROMETHEUS_BASE_URL", "PROMETHEUS_BEARER_TOKEN",
+                  "PROMETHEUS_CA_CERT_PATH", "PROMETHEUS_TOKEN_PATH"):
+            monkeypatch.delenv(k, raising=False)
+        cfg = collector.prometheus_config_from_env({})
+        assert cfg["base_url"] == ""
+        assert cfg["token"] is None
+        assert cfg["insecure_skip_verify"] is False  # TLS verification default on
+        assert cfg["allow_http"] is False
+
+    def test_env_overrides_configmap(self, monkeypatch):
+        monkeypatch.setenv("PROMETHEUS_BASE_URL", "https://env:9090")
+        cfg = collector.prometheus_config_from_env(
+            {"PROMETHEUS_BASE_URL": "https://cm:9090"})
+        assert cfg["base_url"] == "https://env:9090"
+
+    def test_configmap_fallback(self, monkeypatch):
+        monkeypatch.delenv("PROMETHEUS_BASE_URL", raising=False)
+        cfg = collector.prometheus_config_from_env(
+            {"PROMETHEUS_BASE_URL": "https://cm:9090",
+             "PROMETHEUS_TLS_INSECURE_SKIP_VERIFY": "true"})
+        assert cfg["base_url"] == "https://cm:9090"
+        assert cfg["insecure_skip_verify"] is True
+
+    def test_https_enforced_by_client(self):
+        with pytest.raises(ValueError):
+            collector.PrometheusClient(base_url="http://insecure:9090")
+
+    def test_http_allowed_when_opted_in(self):
+        c = collector.PrometheusClient(base_url="http://dev:9090", allow_http=True)
+        assert c is not None  # constructed without raising
