@@ -1,5 +1,6 @@
 """Per-step phase timing probe for large fleets (diagnoses step-time outliers)."""
-import sys, time
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import torch
 from inferno_amd.core.system import System
