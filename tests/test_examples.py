@@ -132,3 +132,21 @@ class TestOfflineCli:
 
         assert cli.cmd_analyze(B()) == 0
         assert "accelerator" in capsys.readouterr().out
+
+
+class TestShippedSystemJson:
+    def test_cli_solves_shipped_example(self, capsys):
+        import json as _json
+
+        from inferno_amd import cli
+
+        class A:
+            spec = "examples/system.json"
+            backend = "cpu"
+            json = True
+
+        assert cli.cmd_solve(A()) == 0
+        out = _json.loads(capsys.readouterr().out)
+        assert len(out["allocations"]) == 6
+        for d in out["allocations"].values():
+            assert d["numReplicas"] >= 1
