@@ -16,9 +16,9 @@ from .build import LIB_PATH, build, needs_build
 
 MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
 # N-bucket thresholds (must match WVA_N_SMALL / WVA_N_MED): cells are
-# dispatched to 64/256/512/1024-thread blocks by batch size and regime — the
-# (large-N) cells' latency sets the sweep wall time, so they get wide blocks,
-# while small cells run one barrier-free wave each.
+# dispatched to 64- and 256-thread blocks by batch size — widths picked by
+# A/B measurement on MI355X (see choose_buckets); small/medium cells run one
+# barrier-free wave each, large-N cells get 4-wave blocks.
 N_SMALL = 512
 N_MED = 2048
 
@@ -26,11 +26,11 @@ N_MED = 2048
 def choose_buckets(batch_n):
     """Partition cells into N-buckets and pick each bucket's block size.
 
-    Wide blocks shorten a single cell's latency (more lanes per chain pass)
-    but cost residency (a 1024-thread block is 1-4 blocks/CU); they pay off
-    only in the LATENCY regime, when the bucket has fewer cells than the
-    chip has resident slots. In the throughput regime (cells >> CUs) smaller
-    blocks win: more cells in flight at the same total work.
+    The natural guess — wide blocks shorten a single cell's latency — is
+    WRONG on CDNA4 for this kernel (measured, profiles/REPORT.md): per-eval
+    cost is dominated by the redundant per-lane work, which costs
+    waves-per-SIMD x instructions, while the serial chain states are cheap
+    fp32-LDS FMAs. Narrow blocks therefore win in BOTH regimes.
 
     Returns [(nt, cell_idx int32 array or None, bucket_max_n, count)].
     """
