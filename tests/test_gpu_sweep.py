@@ -301,3 +301,27 @@ class TestMg1ModeGpu:
             assert set(am) == set(bm)
             for acc in am:
                 assert_alloc_close(am[acc], bm[acc], name, acc)
+
+
+class TestBlockWidthTemplates:
+    """The 512/1024-thread kernel templates are reachable via the
+    INFERNO_NT_* tuning knobs (ops/sweep.py); every width must produce the
+    same results as the CPU golden path."""
+
+    @pytest.mark.parametrize("env,val", [
+        ("INFERNO_NT_LARGE", "512"),
+        ("INFERNO_NT_LARGE", "1024"),
+        ("INFERNO_NT_MED", "256"),
+        ("INFERNO_NT_SMALL", "256"),
+    ])
+    def test_width_override_differential(self, env, val, monkeypatch):
+        monkeypatch.setenv(env, val)
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=12, seed=314)
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map), f"feasibility mismatch for {name}"
+            for acc in a_map:
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
