@@ -135,3 +135,62 @@ class TestShardedGreedy:
             name: (d.accelerator, d.numReplicas) for name, d in result.solution.items()
         }
         assert json.loads(results[0]) == {k: list(v) for k, v in single.items()}
+
+
+class TestShardedSolverWorld3:
+    """Non-power-of-2 world size: 9 servers over 3 ranks (uneven shards are
+    exercised by 4 servers on rank 0 vs 2 on rank 2 with 10 servers)."""
+
+    def test_three_rank_solution_matches_single(self):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        port = 29819
+        procs = [ctx.Process(target=_worker3, args=(r, 3, port, q)) for r in range(3)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(3):
+            rank, payload = q.get(timeout=120)
+            results[rank] = payload
+        for p in procs:
+            p.join(timeout=30)
+            assert p.exitcode == 0
+        assert results[0] == results[1] == results[2]
+
+        from inferno_amd.core.system import System
+        from inferno_amd.engine import SweepEngine
+        from inferno_amd.parallel import ShardedSolver
+        from tests.fixtures import make_spec
+
+        spec = make_spec(n_servers=10, seed=91)
+        system, opt = System.from_spec(spec)
+        result = ShardedSolver(SweepEngine(backend="cpu")).solve(system, opt)
+        single = {
+            name: (d.accelerator, d.numReplicas, round(d.cost, 4))
+            for name, d in result.solution.items()
+        }
+        assert json.loads(results[0]) == {k: list(v) for k, v in single.items()}
+
+
+def _worker3(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+
+    from inferno_amd.core.system import System
+    from inferno_amd.engine import SweepEngine
+    from inferno_amd.parallel import ShardedSolver
+    from tests.fixtures import make_spec
+
+    dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+    try:
+        spec = make_spec(n_servers=10, seed=91)
+        system, opt = System.from_spec(spec)
+        result = ShardedSolver(SweepEngine(backend="cpu")).solve(system, opt)
+        payload = {
+            name: (d.accelerator, d.numReplicas, round(d.cost, 4))
+            for name, d in result.solution.items()
+        }
+        q.put((rank, json.dumps(payload, sort_keys=True)))
+    finally:
+        dist.destroy_process_group()
