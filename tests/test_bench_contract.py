@@ -1,0 +1,60 @@
+"""Driver-contract test: bench.py must emit exactly one parseable JSON line
+with the agreed schema (BASELINE.json metric, value semantics, config block).
+Runs the CPU backend on a tiny fleet so it finishes in seconds without a GPU.
+"""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_bench(*extra):
+    proc = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--backend", "cpu",
+         "--models-per-gpu", "4", "--steps", "2", "--warmup", "1", *extra],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [l for l in proc.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line, got {len(lines)}"
+    return json.loads(lines[0])
+
+
+class TestBenchContract:
+    def test_schema(self):
+        d = run_bench()
+        # required fields of the driver contract
+        for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                    "ms_per_step", "higher_is_better", "scaling",
+                    "vs_baseline", "dtype", "data", "config"):
+            assert key in d, f"missing {key}"
+        assert d["n_gpus"] == 1
+        assert d["steps"] == 2 and d["warmup"] == 1
+        assert d["higher_is_better"] is True
+        assert d["scaling"] == "weak"
+        assert d["unit"] == "allocations/sec"
+        assert d["value"] > 0
+        assert d["ms_per_step"] > 0
+        assert "synthetic" in d["data"]
+        cfg = d["config"]
+        for key in ("model", "global_batch", "parallelism", "reconcile_p50_ms",
+                    "cells_per_step", "analyzer", "solver"):
+            assert key in cfg, f"missing config.{key}"
+        # whole-job aggregate: value = cells_per_step * steps / elapsed
+        assert cfg["cells_per_step"] == 4 * 3  # 4 models x 3 accelerators
+        assert d["value"] == pytest.approx(
+            cfg["cells_per_step"] * d["steps"] / (d["ms_per_step"] * d["steps"] / 1000.0),
+            rel=0.01,
+        )
+
+    def test_mg1_flag(self):
+        d = run_bench("--cpu-analyzer", "mg1")
+        assert d["config"]["analyzer"] == "mg1"
+
+    def test_limited_flag(self):
+        d = run_bench("--limited", "8")
+        assert "greedy limited" in d["config"]["solver"]
