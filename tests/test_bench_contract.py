@@ -58,3 +58,28 @@ class TestBenchContract:
     def test_limited_flag(self):
         d = run_bench("--limited", "8")
         assert "greedy limited" in d["config"]["solver"]
+
+
+class TestBenchDistributedContract:
+    """The driver launches bench.py through torch.distributed.run for N>1 —
+    validate that exact invocation on CPU (gloo, world 2)."""
+
+    def test_torchrun_two_ranks(self):
+        env = dict(os.environ)
+        env["INFERNO_DIST_BACKEND"] = "gloo"
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29921",
+             os.path.join(REPO, "bench.py"), "--gpus", "2", "--backend", "cpu",
+             "--models-per-gpu", "4", "--steps", "2", "--warmup", "1"],
+            capture_output=True, text=True, timeout=600, cwd=REPO, env=env,
+        )
+        assert proc.returncode == 0, proc.stderr[-2000:]
+        lines = [l for l in proc.stdout.splitlines() if l.startswith("{")]
+        assert len(lines) == 1, f"expected one JSON line (rank 0 only): {lines}"
+        d = json.loads(lines[0])
+        assert d["n_gpus"] == 2
+        # weak scaling: 2 ranks x 4 models x 3 accelerators
+        assert d["config"]["cells_per_step"] == 2 * 4 * 3
+        assert "dp2" in d["config"]["parallelism"]
