@@ -876,6 +876,10 @@ extern "C" int wva_sweep_launch_bucket(
       hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,
                          in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
       break;
+    case 128:
+      hipLaunchKernelGGL(wva_sweep_t<128>, dim3(n_blocks), dim3(128), lds, (hipStream_t)stream,
+                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
+      break;
     case 256:
       hipLaunchKernelGGL(wva_sweep_t<256>, dim3(n_blocks), dim3(256), lds, (hipStream_t)stream,
                          in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
@@ -1089,6 +1093,10 @@ static int wva_launch_bucket_on(WvaCtx *c, const WvaBucket &b, hipStream_t s) {
   switch (b.nt) {
     case 64:
       hipLaunchKernelGGL(wva_sweep_t<64>, dim3(b.n_blocks), dim3(64), lds, s, c->in, c->out,
+                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
+      break;
+    case 128:
+      hipLaunchKernelGGL(wva_sweep_t<128>, dim3(b.n_blocks), dim3(128), lds, s, c->in, c->out,
                          b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
       break;
     case 256:
