@@ -311,6 +311,7 @@ class TestBlockWidthTemplates:
     @pytest.mark.parametrize("env,val", [
         ("INFERNO_NT_LARGE", "512"),
         ("INFERNO_NT_LARGE", "1024"),
+        ("INFERNO_NT_MED", "128"),
         ("INFERNO_NT_MED", "256"),
         ("INFERNO_NT_SMALL", "256"),
     ])
