@@ -19,8 +19,8 @@ MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
 # dispatched to 64- and 256-thread blocks by batch size — widths picked by
 # A/B measurement on MI355X (see choose_buckets); small/medium cells run one
 # barrier-free wave each, large-N cells get 4-wave blocks.
-N_SMALL = 512
-N_MED = 2048
+N_SMALL = int(os.environ.get("INFERNO_N_SMALL", "512"))
+N_MED = int(os.environ.get("INFERNO_N_MED", "2048"))
 
 
 def choose_buckets(batch_n):
