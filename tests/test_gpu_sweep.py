@@ -326,3 +326,42 @@ class TestBlockWidthTemplates:
             assert set(a_map) == set(b_map), f"feasibility mismatch for {name}"
             for acc in a_map:
                 assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+
+
+class TestConfig5ShapeDifferential:
+    """BASELINE config 5's 8-variant accelerator ladder (3 SKUs + MI355X
+    TP 2/4/8 + 2 spot tiers) — differentially verify the exact shape the
+    config5 bench sweeps (the plain soaks only cover 1-3 accelerators)."""
+
+    def _fleet(self, seed):
+        from dataclasses import replace
+
+        from inferno_amd.perfmodel import MI355X, accelerator_spec
+        from inferno_amd.utils.synthetic import AMD_ACCELERATORS, make_fleet_spec
+
+        accs = (
+            list(AMD_ACCELERATORS)
+            + [accelerator_spec(MI355X, tp) for tp in (2, 4, 8)]
+            + [
+                replace(AMD_ACCELERATORS[0], name="MI300X-spot", cost=45.0),
+                replace(AMD_ACCELERATORS[2], name="MI355X-spot", cost=70.0),
+            ]
+        )
+        return make_fleet_spec(24, seed=seed, accelerators=accs)
+
+    @pytest.mark.parametrize("seed", [900, 901])
+    def test_eight_variant_differential(self, seed):
+        spec = self._fleet(seed)
+        cpu_sys, _ = System.from_spec(spec)
+        gpu_sys, _ = System.from_spec(self._fleet(seed))
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        n_cells = 0
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map), f"feasibility mismatch for {name}"
+            for acc in a_map:
+                n_cells += 1
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+        assert n_cells >= 24 * 4  # most of the 8-variant grid is feasible
