@@ -274,22 +274,36 @@ class ShardedSolver:
         acc_names = sorted(system.accelerators)
         for name in local_names:
             system.servers[name].all_allocations = {}
-        feasible = cells["feasible"]
-        for i in np.nonzero(feasible)[0]:
-            server = system.servers[local_names[int(cells["cell_server"][i])]]
-            acc_key = acc_names[int(cells["cell_acc_idx"][i])]
-            alloc = Allocation(
-                accelerator="" if cells["zero_empty"][i] else acc_key,
-                num_replicas=int(cells["num_replicas"][i]),
-                batch_size=int(cells["batch"][i]),
-                cost=float(cells["cost"][i]),
-                value=float(cells["value"][i]),
-                itl=float(cells["itl"][i]),
-                ttft=float(cells["ttft"][i]),
-                rho=float(cells["rho"][i]),
-                max_arrv_rate_per_replica=float(cells["max_rate"][i]),
+        idx = np.nonzero(cells["feasible"])[0]
+        if not len(idx):
+            return
+        # convert columns to Python scalars once (tolist) instead of paying
+        # the numpy-scalar box/convert cost per field per cell
+        srv_i = cells["cell_server"][idx].tolist()
+        acc_i = cells["cell_acc_idx"][idx].tolist()
+        zero = cells["zero_empty"][idx].tolist()
+        reps = cells["num_replicas"][idx].tolist()
+        batch = cells["batch"][idx].tolist()
+        cost = cells["cost"][idx].tolist()
+        value = cells["value"][idx].tolist()
+        itl = cells["itl"][idx].tolist()
+        ttft = cells["ttft"][idx].tolist()
+        rho = cells["rho"][idx].tolist()
+        max_rate = cells["max_rate"][idx].tolist()
+        servers = system.servers
+        for k in range(len(idx)):
+            acc_key = acc_names[acc_i[k]]
+            servers[local_names[srv_i[k]]].all_allocations[acc_key] = Allocation(
+                accelerator="" if zero[k] else acc_key,
+                num_replicas=reps[k],
+                batch_size=batch[k],
+                cost=cost[k],
+                value=value[k],
+                itl=itl[k],
+                ttft=ttft[k],
+                rho=rho[k],
+                max_arrv_rate_per_replica=max_rate[k],
             )
-            server.all_allocations[acc_key] = alloc
 
     def _solve_slow(self, system: System, local_names, spec: OptimizerSpec, acc_names):
         # greedy limited mode on GPU: candidates from the cached FastSweep
