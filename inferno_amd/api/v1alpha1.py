@@ -135,6 +135,8 @@ class VariantAutoscaling:
     namespace: str = "default"
     labels: dict[str, str] = field(default_factory=dict)
     generation: int = 1
+    uid: str = ""
+    resourceVersion: str = ""
     ownerReferences: list[dict] = field(default_factory=list)
     deletionTimestamp: Optional[str] = None
     spec: VariantAutoscalingSpec = field(default_factory=VariantAutoscalingSpec)
@@ -202,6 +204,10 @@ def va_to_json(va: VariantAutoscaling) -> dict[str, Any]:
         "labels": dict(va.labels),
         "generation": va.generation,
     }
+    if va.uid:
+        meta["uid"] = va.uid
+    if va.resourceVersion:
+        meta["resourceVersion"] = va.resourceVersion
     if va.ownerReferences:
         meta["ownerReferences"] = va.ownerReferences
     if va.deletionTimestamp:
@@ -305,6 +311,8 @@ def va_from_json(doc: dict[str, Any]) -> VariantAutoscaling:
         namespace=meta.get("namespace", "default"),
         labels=dict(meta.get("labels", {}) or {}),
         generation=int(meta.get("generation", 1)),
+        uid=meta.get("uid", "") or "",
+        resourceVersion=str(meta.get("resourceVersion", "") or ""),
         ownerReferences=list(meta.get("ownerReferences", []) or []),
         deletionTimestamp=meta.get("deletionTimestamp"),
         spec=VariantAutoscalingSpec(

@@ -16,6 +16,13 @@ from typing import Any, Optional, Protocol
 from ..api import v1alpha1 as api
 
 
+class ConflictError(Exception):
+    """Optimistic-concurrency conflict (HTTP 409): the object's
+    resourceVersion moved under us. Mirrors the conflict the reference's
+    ``Status().Update()`` gets from apimachinery (internal/utils/utils.go:91-104
+    retries it with backoff)."""
+
+
 @dataclass
 class Deployment:
     name: str
@@ -46,9 +53,16 @@ class InMemoryKube:
         self.configmaps: dict[tuple[str, str], dict[str, str]] = {}
         self.deployments: dict[tuple[str, str], Deployment] = {}
         self.status_updates: list[dict[str, Any]] = []
+        self._rv_counter = 0
+
+    def _next_rv(self) -> str:
+        self._rv_counter += 1
+        return str(self._rv_counter)
 
     # -- setup helpers -------------------------------------------------
     def add_va(self, va: api.VariantAutoscaling) -> None:
+        if not va.resourceVersion:
+            va.resourceVersion = self._next_rv()
         self.vas[(va.namespace, va.name)] = va
 
     def add_configmap(self, namespace: str, name: str, data: dict[str, str]) -> None:
@@ -60,6 +74,10 @@ class InMemoryKube:
     # -- KubeClient ----------------------------------------------------
     def list_variantautoscalings(self) -> list[api.VariantAutoscaling]:
         return [copy.deepcopy(v) for v in self.vas.values()]
+
+    def get_variantautoscaling(self, namespace: str, name: str) -> Optional[api.VariantAutoscaling]:
+        va = self.vas.get((namespace, name))
+        return copy.deepcopy(va) if va is not None else None
 
     def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]:
         cm = self.configmaps.get((namespace, name))
@@ -73,7 +91,18 @@ class InMemoryKube:
         if key not in self.vas:
             raise KeyError(f"VariantAutoscaling {key} not found")
         stored = self.vas[key]
+        # optimistic concurrency: a stale resourceVersion conflicts instead
+        # of clobbering (apiserver PUT/patch-with-rv semantics)
+        if va.resourceVersion and stored.resourceVersion and (
+            va.resourceVersion != stored.resourceVersion
+        ):
+            raise ConflictError(
+                f"resourceVersion conflict on {key}: "
+                f"have {va.resourceVersion}, stored {stored.resourceVersion}"
+            )
         stored.status = copy.deepcopy(va.status)
+        stored.resourceVersion = self._next_rv()
+        va.resourceVersion = stored.resourceVersion
         self.status_updates.append(api.va_to_json(va)["status"])
 
     def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
@@ -148,6 +177,12 @@ class HttpKube:
             return []
         return [api.va_from_json(item) for item in doc.get("items", [])]
 
+    def get_variantautoscaling(self, namespace: str, name: str) -> Optional[api.VariantAutoscaling]:
+        doc = self._get(
+            f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{namespace}/{api.PLURAL}/{name}"
+        )
+        return None if doc is None else api.va_from_json(doc)
+
     def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]:
         doc = self._get(f"/api/v1/namespaces/{namespace}/configmaps/{name}")
         return None if doc is None else dict(doc.get("data", {}) or {})
@@ -173,16 +208,29 @@ class HttpKube:
             f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/"
             f"{api.PLURAL}/{va.name}/status"
         )
+        meta: dict = {"name": va.name, "namespace": va.namespace}
+        # carry resourceVersion so concurrent writers conflict (409) instead
+        # of last-write-wins, matching the reference's Status().Update()
+        if va.resourceVersion:
+            meta["resourceVersion"] = va.resourceVersion
         body = {
             "apiVersion": api.API_VERSION,
             "kind": api.KIND,
-            "metadata": {"name": va.name, "namespace": va.namespace},
+            "metadata": meta,
             "status": api.va_to_json(va)["status"],
         }
         r = self._client.patch(
             path, json=body, headers={"Content-Type": "application/merge-patch+json"}
         )
+        if r.status_code == 409:
+            raise ConflictError(f"resourceVersion conflict updating {va.namespace}/{va.name}")
         r.raise_for_status()
+        try:
+            new_rv = ((r.json().get("metadata") or {}).get("resourceVersion"))
+            if new_rv:
+                va.resourceVersion = str(new_rv)
+        except ValueError:
+            pass
 
     def watch_events(self, resource_version: str = "", timeout_seconds: int = 60):
         """Stream watch events for VariantAutoscalings (the reference's

@@ -1,6 +1,13 @@
 """Lease-based leader election (coordination.k8s.io/v1), the analogue of
 controller-runtime's leader election used by cmd/main.go:201-219 with
-LeaderElectionID "72dd1cf1.llm-d.ai"."""
+LeaderElectionID "72dd1cf1.llm-d.ai".
+
+Fails CLOSED, matching controller-runtime: on any Lease API error,
+unexpected status, or unreachable coordination API the instance is NOT the
+leader and retries with backoff. The only fail-open path is the explicit
+in-process fake (no HTTP client at all — unit-test double) or running with
+``--no-leader-elect``, in which case no elector is constructed.
+"""
 from __future__ import annotations
 
 import time
@@ -30,10 +37,13 @@ class LeaderElector:
 
     def try_acquire(self) -> bool:
         """Acquire or renew the lease; returns True while we are the leader.
-        Falls open (single-replica assumption) when the Lease API is absent."""
+
+        Error behavior is fail-closed: any API error or exception drops
+        leadership immediately (ref cmd/main.go:201-219 — controller-runtime
+        cancels the manager context when renewal fails)."""
         client = getattr(self.kube, "_client", None)
         if client is None:
-            return True  # in-memory fake: no election
+            return True  # in-memory fake: no election (test double only)
         now = time.time()
         if self._is_leader and now - self._last_renew < self.lease_seconds / 3:
             return True
@@ -55,6 +65,7 @@ class LeaderElector:
                     f"/apis/coordination.k8s.io/v1/namespaces/{self.namespace}/leases",
                     json=body,
                 )
+                # 409 = another replica created it first: not the leader
                 self._is_leader = cr.status_code in (200, 201)
             elif r.status_code == 200:
                 lease = r.json()
@@ -75,13 +86,15 @@ class LeaderElector:
                 if holder == self.identity or expired or not holder:
                     lease["spec"]["holderIdentity"] = self.identity
                     lease["spec"]["renewTime"] = _now()
+                    # PUT carries the fetched resourceVersion: a concurrent
+                    # takeover surfaces as 409 and we lose leadership
                     ur = client.put(self._lease_path(), json=lease)
                     self._is_leader = ur.status_code == 200
                 else:
                     self._is_leader = False
             else:
-                self._is_leader = True  # API error: fail open
+                self._is_leader = False  # API error: fail closed
         except Exception:
-            self._is_leader = True  # no coordination API reachable: fail open
+            self._is_leader = False  # coordination API unreachable: fail closed
         self._last_renew = now
         return self._is_leader

@@ -38,47 +38,21 @@ def gpu_health_probe() -> dict:
         return {"gpu": False, "reason": str(e)}
 
 
-def serve_probes(port: int, state: dict) -> None:
-    """healthz/readyz + prometheus metrics on one HTTP port."""
-    from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
-
-    from prometheus_client import generate_latest
-
-    class Handler(BaseHTTPRequestHandler):
-        def do_GET(self):  # noqa: N802
-            if self.path == "/healthz":
-                self._ok(b"ok")
-            elif self.path == "/readyz":
-                if state.get("ready"):
-                    self._ok(b"ok")
-                else:
-                    self.send_response(503)
-                    self.end_headers()
-            elif self.path == "/metrics":
-                body = generate_latest()
-                self.send_response(200)
-                self.send_header("Content-Type", "text/plain; version=0.0.4")
-                self.end_headers()
-                self.wfile.write(body)
-            else:
-                self.send_response(404)
-                self.end_headers()
-
-        def _ok(self, body: bytes):
-            self.send_response(200)
-            self.end_headers()
-            self.wfile.write(body)
-
-        def log_message(self, *a):  # silence
-            pass
-
-    server = ThreadingHTTPServer(("0.0.0.0", port), Handler)
-    threading.Thread(target=server.serve_forever, daemon=True).start()
-
-
 def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--metrics-port", type=int, default=8443)
+    p.add_argument("--metrics-cert-dir", default=os.environ.get("METRICS_CERT_DIR", ""),
+                   help="dir with tls.crt/tls.key; enables HTTPS + hot reload "
+                        "(ref cmd/main.go:122-155 certwatcher)")
+    p.add_argument("--metrics-client-ca", default=os.environ.get("METRICS_CLIENT_CA", ""),
+                   help="CA bundle; requires verified client certs on /metrics")
+    p.add_argument("--metrics-auth-token-file",
+                   default=os.environ.get("METRICS_AUTH_TOKEN_FILE", ""),
+                   help="static bearer token file protecting /metrics")
+    p.add_argument("--metrics-auth-k8s", action="store_true",
+                   default=os.environ.get("METRICS_AUTH_K8S", "") == "true",
+                   help="delegate /metrics bearer tokens to the TokenReview API "
+                        "(ref cmd/main.go:157-169 authn/authz filter)")
     p.add_argument("--leader-elect", action="store_true", default=True)
     p.add_argument("--no-leader-elect", dest="leader_elect", action="store_false")
     p.add_argument("--configmap-namespace", default=CONFIGMAP_NAMESPACE)
@@ -87,9 +61,26 @@ def main() -> None:
 
     logger = init_logger()
     state = {"ready": False}
-    serve_probes(args.metrics_port, state)
 
     kube = HttpKube()
+
+    from .serving import MetricsAuth, ProbeServer
+
+    auth = MetricsAuth(
+        token_file=args.metrics_auth_token_file or None,
+        kube=kube if args.metrics_auth_k8s else None,
+    )
+    probe_server = ProbeServer(
+        args.metrics_port, state,
+        cert_dir=args.metrics_cert_dir or None,
+        client_ca=args.metrics_client_ca or None,
+        auth=auth,
+    )
+    logger.info(
+        "metrics/probe server started",
+        extra={"kv": {"port": probe_server.port, "tls": probe_server.tls,
+                      "auth": auth.enabled}},
+    )
     cm = kube.get_configmap(args.configmap_namespace, WVA_CONFIG_CM) or {}
     prom_cfg = collector.prometheus_config_from_env(cm)
     prom = collector.PrometheusClient(**prom_cfg)

@@ -22,7 +22,7 @@ from ..core.system import System
 from ..engine import SweepEngine
 from ..parallel import ShardedSolver
 from . import adapters, collector
-from .k8s import Deployment, KubeClient
+from .k8s import ConflictError, Deployment, KubeClient
 from .metrics import MetricsEmitter
 
 CONFIGMAP_NAMESPACE = "workload-variant-autoscaler-system"
@@ -126,6 +126,13 @@ class Reconciler:
         self.configmap_namespace = configmap_namespace
         self.scale_to_zero = scale_to_zero
         self.last_result: Optional[ReconcileResult] = None
+        # GPU->CPU degradation bookkeeping: after a GPU solver failure we run
+        # on the CPU golden, but re-probe the GPU backend every
+        # ``gpu_reprobe_interval`` reconciles so a transient HIP error does
+        # not leave the controller degraded until restart.
+        self._requested_backend = backend
+        self._degraded_ticks = 0
+        self.gpu_reprobe_interval = 10
 
     # ------------------------------------------------------------------
     def read_interval(self) -> float:
@@ -145,8 +152,44 @@ class Reconciler:
         return self.kube.get_configmap(self.configmap_namespace, SERVICE_CLASS_CM)
 
     # ------------------------------------------------------------------
+    def _update_status_with_retry(self, va: api.VariantAutoscaling,
+                                  attempts: int = 3) -> None:
+        """Status write with conflict-refetch-retry, the analogue of the
+        reference's UpdateStatusWithBackoff (internal/utils/utils.go:91-104):
+        on a 409, refetch the live object, graft our computed status onto the
+        fresh resourceVersion and retry."""
+        for i in range(attempts):
+            try:
+                self.kube.update_va_status(va)
+                return
+            except ConflictError:
+                if i == attempts - 1:
+                    raise
+                getter = getattr(self.kube, "get_variantautoscaling", None)
+                fresh = getter(va.namespace, va.name) if getter else None
+                if fresh is None:
+                    raise
+                va.resourceVersion = fresh.resourceVersion
+
+    def _maybe_reprobe_gpu(self) -> None:
+        if not self.engine.degraded_from_gpu:
+            return
+        self._degraded_ticks += 1
+        if self._degraded_ticks < self.gpu_reprobe_interval:
+            return
+        self._degraded_ticks = 0
+        try:
+            probe = SweepEngine(backend="gpu")
+            if probe.backend == "gpu":
+                self.engine = probe
+                self.solver = ShardedSolver(self.engine)
+        except Exception:
+            pass
+
+    # ------------------------------------------------------------------
     def reconcile(self) -> ReconcileResult:
         t_start = time.perf_counter()
+        self._maybe_reprobe_gpu()
         result = ReconcileResult(requeue_after=self.read_interval())
 
         accelerator_cm = self._read_accelerator_cm()
@@ -245,8 +288,10 @@ class Reconciler:
         except Exception as e:
             if self.engine.backend == "gpu":
                 # GPU failure -> CPU reference fallback with SolverDegraded
-                # (SURVEY.md section 5 failure-detection plan)
+                # (SURVEY.md section 5 failure-detection plan); periodically
+                # re-probed by _maybe_reprobe_gpu
                 self.engine = SweepEngine(backend="cpu")
+                self.engine.degraded_from_gpu = True
                 self.solver = ShardedSolver(self.engine)
                 try:
                     shard_result = self.solver.solve(system, opt_spec)
@@ -288,7 +333,7 @@ class Reconciler:
             except Exception as e:
                 result.errors.append(f"{va.name}: metric emission failed: {e}")
             try:
-                self.kube.update_va_status(va)
+                self._update_status_with_retry(va)
             except Exception as e:
                 result.errors.append(f"{va.name}: status update failed: {e}")
                 continue
@@ -307,7 +352,7 @@ class Reconciler:
                 f"Optimization failed: {err}",
             )
             try:
-                self.kube.update_va_status(va)
+                self._update_status_with_retry(va)
             except Exception:
                 pass
         result.errors.append(f"optimization failed: {err}")

@@ -56,6 +56,9 @@ class SweepEngine:
             raise ValueError(f"unknown backend {backend!r}")
         self.backend = backend
         self.device = device
+        # set by the reconciler when this engine is a CPU fallback standing
+        # in for a failed GPU backend (drives periodic GPU re-probe)
+        self.degraded_from_gpu = False
 
     # ------------------------------------------------------------------
     def sweep(self, system: System, server_names: Optional[list[str]] = None) -> EngineStats:

@@ -431,11 +431,12 @@ class TestProbeServer:
     def test_probe_endpoints(self):
         import urllib.request
 
-        from inferno_amd.controller.main import gpu_health_probe, serve_probes
+        from inferno_amd.controller.main import gpu_health_probe
+        from inferno_amd.controller.serving import ProbeServer
 
         state = {"ready": False}
-        serve_probes(18473, state)
-        base = "http://127.0.0.1:18473"
+        srv = ProbeServer(0, state, bind="127.0.0.1")
+        base = f"http://127.0.0.1:{srv.port}"
 
         def get(path):
             try:
@@ -532,3 +533,90 @@ class TestPrometheusConfigFromEnv:
     def test_http_allowed_when_opted_in(self):
         c = collector.PrometheusClient(base_url="http://dev:9090", allow_http=True)
         assert c is not None  # constructed without raising
+
+
+class TestConflictRetry:
+    """Status writes retry through resourceVersion conflicts
+    (UpdateStatusWithBackoff analogue, internal/utils/utils.go:91-104)."""
+
+    def test_conflict_then_refetch_succeeds(self):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=6.0)
+        real_update = kube.update_va_status
+        calls = {"n": 0}
+
+        def flaky_update(va):
+            calls["n"] += 1
+            if calls["n"] == 1:
+                # another writer bumped the object under us
+                from inferno_amd.controller.k8s import ConflictError
+
+                raise ConflictError("stale rv")
+            real_update(va)
+
+        kube.update_va_status = flaky_update
+        result = rec.reconcile()
+        assert result.processed == 1
+        assert calls["n"] == 2  # one conflict, one successful retry
+        assert not any("status update failed" in e for e in result.errors)
+
+    def test_persistent_conflict_surfaces_error(self):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=6.0)
+
+        def always_conflict(va):
+            from inferno_amd.controller.k8s import ConflictError
+
+            raise ConflictError("stale rv")
+
+        kube.update_va_status = always_conflict
+        result = rec.reconcile()
+        assert result.processed == 0
+        assert any("status update failed" in e for e in result.errors)
+
+
+class TestGpuReprobe:
+    """After a GPU solver failure the CPU fallback is periodically re-probed
+    instead of degrading permanently (ADVICE r1 low #3)."""
+
+    def test_reprobe_counter_and_attempt(self, monkeypatch):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=2.0)
+        rec.engine.degraded_from_gpu = True
+        rec.gpu_reprobe_interval = 3
+        probes = {"n": 0}
+
+        import inferno_amd.controller.reconciler as rmod
+
+        class FakeEngine:
+            backend = "cpu"  # probe "fails": stays cpu -> no swap
+            degraded_from_gpu = False
+
+            def __init__(self, backend="auto", device="cuda"):
+                probes["n"] += 1
+
+        monkeypatch.setattr(rmod, "SweepEngine", FakeEngine)
+        for _ in range(7):
+            rec.reconcile()
+        # ticks 3 and 6 triggered a probe
+        assert probes["n"] == 2
+        assert rec.engine.degraded_from_gpu is True  # probe failed, still degraded
+
+    def test_successful_reprobe_restores_gpu(self, monkeypatch):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=2.0)
+        rec.engine.degraded_from_gpu = True
+        rec.gpu_reprobe_interval = 1
+
+        import inferno_amd.controller.reconciler as rmod
+
+        real_engine = type(rec.engine)
+
+        class GpuOkEngine:
+            def __new__(cls, backend="auto", device="cuda"):
+                # build a real CPU engine but label it as a healthy GPU probe
+                eng = real_engine(backend="cpu")
+                eng.backend_probe = True
+                eng.backend = "gpu"
+                return eng
+
+        monkeypatch.setattr(rmod, "SweepEngine", GpuOkEngine)
+        rec._maybe_reprobe_gpu()
+        assert rec.engine.backend == "gpu"
+        assert rec.engine.degraded_from_gpu is False

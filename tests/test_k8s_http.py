@@ -243,3 +243,124 @@ class TestConfigMapWatch:
             timeout_seconds=1))
         assert events == [("MODIFIED", "service-classes-config"),
                           ("ADDED", "accelerator-unit-costs")]
+
+
+class TestLeaderFailClosed:
+    """Leader election must fail CLOSED on API errors (ADVICE r1 medium;
+    ref cmd/main.go:201-219 controller-runtime semantics)."""
+
+    def test_api_error_drops_leadership(self):
+        def handler(req):
+            return httpx.Response(500, json={})
+
+        el = LeaderElector(make_kube(handler), "lease", "ns", "me")
+        assert el.try_acquire() is False
+
+    def test_unreachable_api_drops_leadership(self):
+        def handler(req):
+            raise httpx.ConnectError("coordination API down")
+
+        el = LeaderElector(make_kube(handler), "lease", "ns", "me")
+        assert el.try_acquire() is False
+
+    def test_renew_conflict_loses_leadership(self):
+        """A 409 on the renew PUT (concurrent takeover with a newer
+        resourceVersion) must drop leadership."""
+        def handler(req):
+            if req.method == "GET":
+                return httpx.Response(200, json={
+                    "metadata": {"resourceVersion": "5"},
+                    "spec": {"holderIdentity": "me",
+                             "renewTime": "2020-01-01T00:00:00.0Z",
+                             "leaseDurationSeconds": 15}})
+            if req.method == "PUT":
+                return httpx.Response(409, json={})
+            return httpx.Response(500)
+
+        el = LeaderElector(make_kube(handler), "lease", "ns", "me")
+        assert el.try_acquire() is False
+
+    def test_error_then_recovery_reacquires(self):
+        state = {"fail": True}
+
+        def handler(req):
+            if state["fail"]:
+                return httpx.Response(503)
+            if req.method == "GET":
+                return httpx.Response(404)
+            if req.method == "POST":
+                return httpx.Response(201, json={})
+            return httpx.Response(500)
+
+        el = LeaderElector(make_kube(handler), "lease", "ns", "me")
+        assert el.try_acquire() is False
+        state["fail"] = False
+        assert el.try_acquire() is True
+
+    def test_create_conflict_not_leader(self):
+        """POST 409 (another replica created the Lease first) => not leader."""
+        def handler(req):
+            if req.method == "GET":
+                return httpx.Response(404)
+            if req.method == "POST":
+                return httpx.Response(409, json={})
+            return httpx.Response(500)
+
+        el = LeaderElector(make_kube(handler), "lease", "ns", "me")
+        assert el.try_acquire() is False
+
+
+class TestStatusOptimisticConcurrency:
+    """resourceVersion carried in status patches; 409 surfaces as
+    ConflictError (ref internal/utils/utils.go:91-104)."""
+
+    def test_patch_carries_resource_version(self):
+        seen = {}
+
+        def handler(req):
+            seen["body"] = json.loads(req.content)
+            return httpx.Response(200, json={"metadata": {"resourceVersion": "43"}})
+
+        kube = make_kube(handler)
+        va = api.VariantAutoscaling(name="v1", namespace="ns", resourceVersion="42")
+        kube.update_va_status(va)
+        assert seen["body"]["metadata"]["resourceVersion"] == "42"
+        # server-returned rv adopted for the next write
+        assert va.resourceVersion == "43"
+
+    def test_conflict_raises(self):
+        from inferno_amd.controller.k8s import ConflictError
+
+        def handler(req):
+            return httpx.Response(409, json={})
+
+        kube = make_kube(handler)
+        va = api.VariantAutoscaling(name="v1", namespace="ns", resourceVersion="42")
+        import pytest
+
+        with pytest.raises(ConflictError):
+            kube.update_va_status(va)
+
+    def test_get_variantautoscaling(self):
+        def handler(req):
+            assert req.url.path == "/apis/llmd.ai/v1alpha1/namespaces/ns/variantautoscalings/v1"
+            return httpx.Response(200, json={
+                "metadata": {"name": "v1", "namespace": "ns", "resourceVersion": "7"},
+                "spec": {"modelID": "m"}})
+
+        va = make_kube(handler).get_variantautoscaling("ns", "v1")
+        assert va.name == "v1" and va.resourceVersion == "7"
+
+    def test_inmemory_conflict_and_rv_bump(self):
+        from inferno_amd.controller.k8s import ConflictError, InMemoryKube
+
+        kube = InMemoryKube()
+        va = api.VariantAutoscaling(name="v1", namespace="ns")
+        kube.add_va(va)
+        fresh = kube.get_variantautoscaling("ns", "v1")
+        stale = kube.get_variantautoscaling("ns", "v1")
+        kube.update_va_status(fresh)  # bumps stored rv
+        import pytest
+
+        with pytest.raises(ConflictError):
+            kube.update_va_status(stale)
