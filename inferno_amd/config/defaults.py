@@ -16,10 +16,10 @@ MAX_QUEUE_TO_BATCH_RATIO = 10
 # Accelerator transition penalty factor (defaults.go:22)
 ACCEL_PENALTY_FACTOR = 0.1
 
-# Hard cap on queueing-chain batch states (max batch size N per cell).
-# The reference has no cap (its chain is an O(11*N) sequential recurrence);
-# this build caps N at the HIP kernel's LDS budget (WVA_MAX_N) and applies
-# the same cap on every backend so CPU and GPU sweeps agree exactly.
+# LDS-resident chain-state threshold (max batch size N per cell that fits
+# the HIP kernel's LDS geometry, WVA_MAX_N). N is NOT capped — matching the
+# reference's uncapped sizing (allocation.go:80-86) — cells above this spill
+# their chain geometry to a global-memory slab on the GPU backend.
 MAX_BATCH_STATES = 8192
 
 # Service-class defaults (defaults.go:24-33)

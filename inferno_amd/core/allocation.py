@@ -23,7 +23,6 @@ from ..analyzer import (
 )
 from ..config import (
     ACCEL_PENALTY_FACTOR,
-    MAX_BATCH_STATES,
     MAX_QUEUE_TO_BATCH_RATIO,
     AllocationData,
 )
@@ -223,7 +222,8 @@ def create_allocation(system, server_name: str, acc_name: str) -> Optional[Alloc
         N = server.max_batch_size
     else:
         N = max(perf.maxBatchSize * perf.atTokens // K, 1)
-    N = min(N, MAX_BATCH_STATES)
+    # N is uncapped, matching the reference (allocation.go:80-86); the GPU
+    # sweep spills chain geometry for N > 8192 to global memory
     max_queue = N * MAX_QUEUE_TO_BATCH_RATIO
 
     cfg = Configuration(

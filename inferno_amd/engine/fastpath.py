@@ -21,7 +21,6 @@ from typing import Optional
 
 import numpy as np
 
-from ..config import MAX_BATCH_STATES
 from ..core import create_allocation
 from ..core.system import System
 from .snapshot import FLAG_CUR_EMPTY, FLAG_CUR_SAME, FLAG_HAS_CUR
@@ -170,10 +169,11 @@ class FastSweep:
                  * self._stat["at_tokens"].astype(np.int64)) // safe_k,
                 1,
             )
+            # uncapped like the reference (allocation.go:80-86); huge-N cells
+            # dispatch to the GMEM spill bucket in choose_buckets
             base_batch = np.where(
                 self._stat["override"] > 0, self._stat["override"], n_scaled
-            )
-            base_batch = np.minimum(base_batch, MAX_BATCH_STATES).astype(np.int32)
+            ).astype(np.int32)
             pmb = np.where(
                 self._stat["override"] > 0, self._stat["override"],
                 self._stat["perf_max_batch_cfg"],
@@ -276,17 +276,19 @@ class FastSweep:
     def _buckets_for(self, batch_n: np.ndarray):
         import torch
 
-        from ..ops.sweep import choose_buckets
+        from ..ops.sweep import alloc_gmem_slabs, choose_buckets
 
         st = self._gpu
         key = batch_n.tobytes()
         if st["bucket_key"] == key:
             return st["buckets"]
-        buckets = [
-            (nt, torch.from_numpy(ids).to(self.device) if ids is not None else None,
-             bmax, count)
-            for nt, ids, bmax, count in choose_buckets(batch_n)
-        ]
+        buckets = []
+        for nt, ids, bmax, count, gmem in choose_buckets(batch_n):
+            ids_t = torch.from_numpy(ids).to(self.device) if ids is not None else None
+            slabs = (
+                alloc_gmem_slabs(count, bmax, nt, self.device) if gmem else (None, None)
+            )
+            buckets.append((nt, ids_t, bmax, count, slabs))
         st["bucket_key"] = key
         st["buckets"] = buckets
         return buckets
@@ -356,11 +358,19 @@ class FastSweep:
             *[int(b[1].numel()) if b[1] is not None else self.n_cells for b in buckets]
         )
         mxn = (ctypes.c_int * n)(*[b[2] for b in buckets])
+        g_invs = (ctypes.c_void_p * n)(
+            *[ctypes.c_void_p(b[4][0].data_ptr()) if b[4][0] is not None else None
+              for b in buckets]
+        )
+        g_anchors = (ctypes.c_void_p * n)(
+            *[ctypes.c_void_p(b[4][1].data_ptr()) if b[4][1] is not None else None
+              for b in buckets]
+        )
         mode = 1 if getattr(self.system, "analyzer_mode", "mm1k") == "mg1" else 0
         cv2 = float(getattr(self.system, "analyzer_cv2", 1.0))
         rc = st["ctx_lib"].wva_ctx_set_buckets(
             st["ctx"], ctypes.c_int(n), nts, ids, nbl, mxn,
-            ctypes.c_int(mode), ctypes.c_float(cv2),
+            ctypes.c_int(mode), ctypes.c_float(cv2), g_invs, g_anchors,
         )
         if rc != 0:
             from ..ops.sweep import HipKernelError

@@ -12,7 +12,6 @@ from dataclasses import dataclass
 
 import numpy as np
 
-from ..config import MAX_BATCH_STATES as MAX_SWEEP_N
 from ..core.system import System
 
 FLAG_CUR_SAME = 1
@@ -36,12 +35,14 @@ class CellSnapshot:
 
 
 def compute_batch_size(server, perf, K: int) -> int:
-    """N from server override or scaled perf data (ref allocation.go:80-86)."""
+    """N from server override or scaled perf data (ref allocation.go:80-86).
+    Uncapped like the reference; N > 8192 cells use the GPU's global-memory
+    geometry spill path."""
     if server.max_batch_size > 0:
         n = server.max_batch_size
     else:
         n = max(perf.maxBatchSize * perf.atTokens // K, 1)
-    return min(n, MAX_SWEEP_N)
+    return n
 
 
 def build_cell_snapshot(system: System, server_names: list[str] | None = None) -> CellSnapshot:

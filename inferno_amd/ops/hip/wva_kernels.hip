@@ -47,9 +47,15 @@
 #include <stdint.h>
 
 #define WVA_WAVE 64
-// max supported batch size per cell (bounds the LDS chain geometry:
-// chunk*NT fp32 reciprocals + NT*ksub fp64 anchors + a 40-double header)
+// max LDS-resident batch size per cell (bounds the LDS chain geometry:
+// chunk*NT fp32 reciprocals + NT*ksub fp64 anchors + a 40-double header).
+// Cells with N above this spill their chain geometry to a per-block global
+// memory slab (GMEM=true instantiation) — the reference's N is uncapped
+// (allocation.go:80-86), so no batch size may change the computed allocation.
 #define WVA_MAX_N 8192
+// sanity guard for the global-memory spill path (absurd profiles fail loudly
+// instead of exhausting HBM: 4M states/cell = ~16 MB geometry per cell)
+#define WVA_HUGE_MAX_N (1 << 22)
 // N-bucket thresholds (host mirrors these in ops/sweep.py)
 #define WVA_N_SMALL 512
 #define WVA_N_MED 2048
@@ -559,10 +565,17 @@ __device__ void dual_bisect(double lam_min, double lam_max, float t_ttft, float 
 // N-bucket (nullptr = identity).
 // dynamic LDS: S[0..maxN] prefix, then scratch[2*NT/64] for reductions.
 // ---------------------------------------------------------------------------
-template <int NT>
+// GMEM=false: chain geometry in LDS (hot path, N <= WVA_MAX_N).
+// GMEM=true: geometry in a per-block global-memory slab (g_inv/g_anchor,
+// strided by the bucket's max_n) — same algorithm, same arithmetic order,
+// bit-identical results; used for the rare huge-N cells so the sweep stays
+// uncapped like the reference. The slab layout mirrors the LDS layout
+// (chunk-transposed reciprocals), which is also the coalesced global layout.
+template <int NT, bool GMEM>
 __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut out, int n_blocks,
                                                   const int *cell_ids, int max_n,
-                                                  int analyzer_mode, float cv2) {
+                                                  int analyzer_mode, float cv2,
+                                                  float *g_inv, double *g_anchor) {
   extern __shared__ double smem[];
   if ((int)blockIdx.x >= n_blocks) return;
   const int cell = cell_ids ? cell_ids[blockIdx.x] : (int)blockIdx.x;
@@ -650,9 +663,19 @@ __global__ void __launch_bounds__(NT, 4) wva_sweep_t(WvaCellsIn in, WvaCellsOut 
   // (skipped entirely in M/G/1 mode — its evaluations are closed-form)
   const int chunk = (N + NT - 1) / NT;
   const int ksub = (chunk + WVA_SUB - 1) / WVA_SUB;
-  // LDS partition (S points at the dynamic smem base; scratch precedes)
-  float *inv_s_t = (float *)S;                // chunk*NT floats
-  double *S_anchor = S + (chunk * NT + 1) / 2;  // NT*ksub doubles (8B aligned)
+  float *inv_s_t;
+  double *S_anchor;
+  if constexpr (GMEM) {
+    // per-block slab in global memory, strided by the bucket's max geometry
+    const int chunk_max = (max_n + NT - 1) / NT;
+    const int ksub_max = (chunk_max + WVA_SUB - 1) / WVA_SUB;
+    inv_s_t = g_inv + (size_t)blockIdx.x * ((size_t)chunk_max * NT);
+    S_anchor = g_anchor + (size_t)blockIdx.x * ((size_t)NT * ksub_max);
+  } else {
+    // LDS partition (S points at the dynamic smem base; scratch precedes)
+    inv_s_t = (float *)S;                 // chunk*NT floats
+    S_anchor = S + (chunk * NT + 1) / 2;  // NT*ksub doubles (8B aligned)
+  }
   double *total_slot = scratch + 32;
 
   const int n0 = tid * chunk + 1;
@@ -849,6 +872,46 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
 // ---------------------------------------------------------------------------
 // C ABI launchers
 // ---------------------------------------------------------------------------
+// runtime (nt, gmem) -> template dispatch. GMEM instantiations exist for the
+// two block widths the host's huge-bucket policy uses (256, 1024).
+static int wva_sweep_dispatch(int n_blocks, int max_n, int nt, const int *cell_ids,
+                              int analyzer_mode, float cv2, hipStream_t s,
+                              const WvaCellsIn &in, const WvaCellsOut &out, float *g_inv,
+                              double *g_anchor) {
+  if (n_blocks <= 0) return 0;
+  const bool gmem = g_inv != nullptr && g_anchor != nullptr;
+  if (max_n < 1) return -2;
+  if (!gmem && max_n > WVA_MAX_N) return -2;
+  if (gmem && max_n > WVA_HUGE_MAX_N) return -4;
+  const int chunk = (max_n + nt - 1) / nt;
+  const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
+  // header(40) + inv_s floats (chunk*nt/2 doubles, rounded up) + anchors;
+  // the spill path keeps only the header in LDS
+  size_t lds = gmem ? (size_t)40 * sizeof(double)
+                    : (size_t)(40 + (chunk * nt + 1) / 2 + (size_t)nt * ksub) * sizeof(double);
+#define WVA_LAUNCH(NTV, GM)                                                                 \
+  hipLaunchKernelGGL((wva_sweep_t<NTV, GM>), dim3(n_blocks), dim3(NTV), lds, s, in, out,    \
+                     n_blocks, cell_ids, max_n, analyzer_mode, cv2, g_inv, g_anchor)
+  if (gmem) {
+    switch (nt) {
+      case 256: WVA_LAUNCH(256, true); break;
+      case 1024: WVA_LAUNCH(1024, true); break;
+      default: return -3;
+    }
+  } else {
+    switch (nt) {
+      case 64: WVA_LAUNCH(64, false); break;
+      case 128: WVA_LAUNCH(128, false); break;
+      case 256: WVA_LAUNCH(256, false); break;
+      case 512: WVA_LAUNCH(512, false); break;
+      case 1024: WVA_LAUNCH(1024, false); break;
+      default: return -3;
+    }
+  }
+#undef WVA_LAUNCH
+  return (int)hipGetLastError();
+}
+
 extern "C" int wva_sweep_launch_bucket(
     int n_blocks, int max_n, int nt, const int *cell_ids, int analyzer_mode, float cv2,
     void *stream,
@@ -858,45 +921,15 @@ extern "C" int wva_sweep_launch_bucket(
     const float *arrival_rate, const float *t_itl, const float *t_ttft, const float *t_tps,
     const float *acc_cost, const float *cur_cost,
     uint8_t *feasible, uint8_t *zero_empty, int *num_replicas, int *batch, float *cost,
-    float *value, float *itl, float *ttft, float *rho, float *max_rate) {
-  if (n_blocks <= 0) return 0;
-  if (max_n < 1 || max_n > WVA_MAX_N) return -2;
+    float *value, float *itl, float *ttft, float *rho, float *max_rate,
+    void *g_inv, void *g_anchor) {
   WvaCellsIn in = {in_tok, out_tok, batch_n, min_replicas, perf_max_batch, cur_replicas, flags,
                    alpha, beta, gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
                    cur_cost};
   WvaCellsOut out = {feasible, zero_empty, num_replicas, batch, cost, value, itl, ttft, rho,
                      max_rate};
-  const int chunk = (max_n + nt - 1) / nt;
-  const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
-  // header(40) + inv_s floats (chunk*nt/2 doubles, rounded up) + anchors
-  size_t lds =
-      (size_t)(40 + (chunk * nt + 1) / 2 + (size_t)nt * ksub) * sizeof(double);
-  switch (nt) {
-    case 64:
-      hipLaunchKernelGGL(wva_sweep_t<64>, dim3(n_blocks), dim3(64), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
-      break;
-    case 128:
-      hipLaunchKernelGGL(wva_sweep_t<128>, dim3(n_blocks), dim3(128), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
-      break;
-    case 256:
-      hipLaunchKernelGGL(wva_sweep_t<256>, dim3(n_blocks), dim3(256), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
-      break;
-    case 512:
-      hipLaunchKernelGGL(wva_sweep_t<512>, dim3(n_blocks), dim3(512), lds, (hipStream_t)stream,
-                         in, out, n_blocks, cell_ids, max_n, analyzer_mode, cv2);
-      break;
-    case 1024:
-      hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(n_blocks), dim3(1024), lds,
-                         (hipStream_t)stream, in, out, n_blocks, cell_ids, max_n,
-                         analyzer_mode, cv2);
-      break;
-    default:
-      return -3;
-  }
-  return (int)hipGetLastError();
+  return wva_sweep_dispatch(n_blocks, max_n, nt, cell_ids, analyzer_mode, cv2,
+                            (hipStream_t)stream, in, out, (float *)g_inv, (double *)g_anchor);
 }
 
 // single-bucket convenience wrapper (whole sweep with one block size)
@@ -915,7 +948,7 @@ extern "C" int wva_sweep_launch(
                                  min_replicas, perf_max_batch, cur_replicas, flags, alpha, beta,
                                  gamma, delta, arrival_rate, t_itl, t_ttft, t_tps, acc_cost,
                                  cur_cost, feasible, zero_empty, num_replicas, batch, cost,
-                                 value, itl, ttft, rho, max_rate);
+                                 value, itl, ttft, rho, max_rate, nullptr, nullptr);
 }
 
 extern "C" int wva_argmin_launch(int n_servers, void *stream, const float *value,
@@ -963,13 +996,15 @@ extern "C" __global__ void __launch_bounds__(256) wva_gather(
 // three streams, argmin, gather, D2H, sync) — no per-step host dispatch
 // beyond wva_reconcile(ctx).
 // ---------------------------------------------------------------------------
-#define WVA_MAX_BUCKETS 3
+#define WVA_MAX_BUCKETS 4
 
 struct WvaBucket {
   int nt;
   const int *cell_ids;  // nullptr = identity
   int n_blocks;
   int max_n;
+  float *g_inv;      // non-null: global-memory geometry slab (huge-N spill)
+  double *g_anchor;
 };
 
 struct WvaCtx {
@@ -994,8 +1029,8 @@ struct WvaCtx {
   void *pin_out_i;    // host pinned, 4 x n_srv i32
   WvaBucket buckets[WVA_MAX_BUCKETS];
   int n_buckets;
-  hipStream_t s0, s1, s2;
-  hipEvent_t e_up, e_b1, e_b2;
+  hipStream_t s0, s1, s2, s3;
+  hipEvent_t e_up, e_b1, e_b2, e_b3;
 };
 
 // pointer-slot order for wva_ctx_create (host mirrors in ops/sweep.py):
@@ -1060,9 +1095,11 @@ extern "C" void *wva_ctx_create(int n_cells, int n_srv, void **p) {
   if (hipStreamCreateWithFlags(&c->s0, hipStreamNonBlocking) != hipSuccess ||
       hipStreamCreateWithFlags(&c->s1, hipStreamNonBlocking) != hipSuccess ||
       hipStreamCreateWithFlags(&c->s2, hipStreamNonBlocking) != hipSuccess ||
+      hipStreamCreateWithFlags(&c->s3, hipStreamNonBlocking) != hipSuccess ||
       hipEventCreateWithFlags(&c->e_up, hipEventDisableTiming) != hipSuccess ||
       hipEventCreateWithFlags(&c->e_b1, hipEventDisableTiming) != hipSuccess ||
-      hipEventCreateWithFlags(&c->e_b2, hipEventDisableTiming) != hipSuccess) {
+      hipEventCreateWithFlags(&c->e_b2, hipEventDisableTiming) != hipSuccess ||
+      hipEventCreateWithFlags(&c->e_b3, hipEventDisableTiming) != hipSuccess) {
     delete c;
     return nullptr;
   }
@@ -1071,7 +1108,8 @@ extern "C" void *wva_ctx_create(int n_cells, int n_srv, void **p) {
 
 extern "C" int wva_ctx_set_buckets(void *ctx, int n_buckets, const int *nts,
                                    const void **cell_ids, const int *n_blocks,
-                                   const int *max_ns, int analyzer_mode, float cv2) {
+                                   const int *max_ns, int analyzer_mode, float cv2,
+                                   const void **g_invs, const void **g_anchors) {
   WvaCtx *c = (WvaCtx *)ctx;
   if (n_buckets < 0 || n_buckets > WVA_MAX_BUCKETS) return -1;
   c->n_buckets = n_buckets;
@@ -1080,6 +1118,8 @@ extern "C" int wva_ctx_set_buckets(void *ctx, int n_buckets, const int *nts,
     c->buckets[i].cell_ids = (const int *)cell_ids[i];
     c->buckets[i].n_blocks = n_blocks[i];
     c->buckets[i].max_n = max_ns[i];
+    c->buckets[i].g_inv = g_invs ? (float *)g_invs[i] : nullptr;
+    c->buckets[i].g_anchor = g_anchors ? (double *)g_anchors[i] : nullptr;
   }
   c->analyzer_mode = analyzer_mode;
   c->cv2 = cv2;
@@ -1087,34 +1127,8 @@ extern "C" int wva_ctx_set_buckets(void *ctx, int n_buckets, const int *nts,
 }
 
 static int wva_launch_bucket_on(WvaCtx *c, const WvaBucket &b, hipStream_t s) {
-  const int chunk = (b.max_n + b.nt - 1) / b.nt;
-  const int ksub = (chunk + 32 - 1) / 32;
-  size_t lds = (size_t)(40 + (chunk * b.nt + 1) / 2 + (size_t)b.nt * ksub) * sizeof(double);
-  switch (b.nt) {
-    case 64:
-      hipLaunchKernelGGL(wva_sweep_t<64>, dim3(b.n_blocks), dim3(64), lds, s, c->in, c->out,
-                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
-      break;
-    case 128:
-      hipLaunchKernelGGL(wva_sweep_t<128>, dim3(b.n_blocks), dim3(128), lds, s, c->in, c->out,
-                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
-      break;
-    case 256:
-      hipLaunchKernelGGL(wva_sweep_t<256>, dim3(b.n_blocks), dim3(256), lds, s, c->in, c->out,
-                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
-      break;
-    case 512:
-      hipLaunchKernelGGL(wva_sweep_t<512>, dim3(b.n_blocks), dim3(512), lds, s, c->in, c->out,
-                         b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
-      break;
-    case 1024:
-      hipLaunchKernelGGL(wva_sweep_t<1024>, dim3(b.n_blocks), dim3(1024), lds, s, c->in,
-                         c->out, b.n_blocks, b.cell_ids, b.max_n, c->analyzer_mode, c->cv2);
-      break;
-    default:
-      return -3;
-  }
-  return (int)hipGetLastError();
+  return wva_sweep_dispatch(b.n_blocks, b.max_n, b.nt, b.cell_ids, c->analyzer_mode, c->cv2, s,
+                            c->in, c->out, b.g_inv, b.g_anchor);
 }
 
 extern "C" int wva_reconcile(void *ctx) {
@@ -1129,18 +1143,18 @@ extern "C" int wva_reconcile(void *ctx) {
   if (err != hipSuccess) return (int)err;
   if (hipEventRecord(c->e_up, c->s0) != hipSuccess) return -10;
 
-  // 2. bucket launches: bucket 0 on s0; others overlap on s1/s2 after upload
-  hipStream_t side[2] = {c->s1, c->s2};
+  // 2. bucket launches: bucket 0 on s0; others overlap on s1/s2/s3 after upload
+  hipStream_t side[3] = {c->s1, c->s2, c->s3};
+  hipEvent_t bev[3] = {c->e_b1, c->e_b2, c->e_b3};
   for (int i = 0; i < c->n_buckets; ++i) {
     hipStream_t s = (i == 0) ? c->s0 : side[i - 1];
     if (i > 0 && hipStreamWaitEvent(s, c->e_up, 0) != hipSuccess) return -11;
     int rc = wva_launch_bucket_on(c, c->buckets[i], s);
     if (rc != 0) return rc;
-    if (i == 1 && hipEventRecord(c->e_b1, s) != hipSuccess) return -12;
-    if (i == 2 && hipEventRecord(c->e_b2, s) != hipSuccess) return -13;
+    if (i > 0 && hipEventRecord(bev[i - 1], s) != hipSuccess) return -12;
   }
-  if (c->n_buckets > 1 && hipStreamWaitEvent(c->s0, c->e_b1, 0) != hipSuccess) return -14;
-  if (c->n_buckets > 2 && hipStreamWaitEvent(c->s0, c->e_b2, 0) != hipSuccess) return -15;
+  for (int i = 1; i < c->n_buckets; ++i)
+    if (hipStreamWaitEvent(c->s0, bev[i - 1], 0) != hipSuccess) return -14;
 
   // 3. argmin + gather on s0
   hipLaunchKernelGGL(wva_argmin, dim3(c->n_srv), dim3(WVA_WAVE), 0, c->s0, c->out.value,
@@ -1168,8 +1182,10 @@ extern "C" void wva_ctx_destroy(void *ctx) {
   (void)hipStreamDestroy(c->s0);
   (void)hipStreamDestroy(c->s1);
   (void)hipStreamDestroy(c->s2);
+  (void)hipStreamDestroy(c->s3);
   (void)hipEventDestroy(c->e_up);
   (void)hipEventDestroy(c->e_b1);
   (void)hipEventDestroy(c->e_b2);
+  (void)hipEventDestroy(c->e_b3);
   delete c;
 }

@@ -14,7 +14,8 @@ from typing import Optional
 
 from .build import LIB_PATH, build, needs_build
 
-MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip
+MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip (LDS-resident limit)
+HUGE_MAX_N = 1 << 22  # must match WVA_HUGE_MAX_N (global-memory spill limit)
 # N-bucket thresholds (must match WVA_N_SMALL / WVA_N_MED): cells are
 # dispatched to 64- and 256-thread blocks by batch size — widths picked by
 # A/B measurement on MI355X (see choose_buckets); small/medium cells run one
@@ -32,7 +33,12 @@ def choose_buckets(batch_n):
     waves-per-SIMD x instructions, while the serial chain states are cheap
     fp32-LDS FMAs. Narrow blocks therefore win in BOTH regimes.
 
-    Returns [(nt, cell_idx int32 array or None, bucket_max_n, count)].
+    Cells with N above MAX_N (the LDS budget) go to a fourth, global-memory
+    bucket: their chain geometry lives in a per-block HBM slab instead of LDS
+    (GMEM kernel instantiation) so the sweep is uncapped like the reference
+    (allocation.go:80-86).
+
+    Returns [(nt, cell_idx int32 array or None, bucket_max_n, count, gmem)].
     """
     import numpy as np
 
@@ -40,13 +46,26 @@ def choose_buckets(batch_n):
     masks = [
         (64, batch_n <= N_SMALL),
         (256, (batch_n > N_SMALL) & (batch_n <= N_MED)),
-        (1024, batch_n > N_MED),
+        (1024, (batch_n > N_MED) & (batch_n <= MAX_N)),
+        ("huge", batch_n > MAX_N),
     ]
     out = []
     for base_nt, mask in masks:
         idx = np.nonzero(mask)[0]
         count = len(idx)
         if count == 0:
+            continue
+        if base_nt == "huge":
+            bmax = int(batch_n[mask].max())
+            if bmax > HUGE_MAX_N:
+                raise HipKernelError(
+                    f"batch size {bmax} exceeds the global-memory sweep limit "
+                    f"{HUGE_MAX_N} (per-cell geometry would be "
+                    f"{bmax * 4 / 1e6:.0f}+ MB)"
+                )
+            nt = int(os.environ.get("INFERNO_NT_HUGE", "256"))
+            ids = None if count == n else idx.astype(np.int32)
+            out.append((nt, ids, bmax, count, True))
             continue
         # Measured on MI355X (profiles/REPORT.md, 512-model fleet A/B): the
         # per-eval cost of a bisection step is dominated by the REDUNDANT
@@ -65,8 +84,21 @@ def choose_buckets(batch_n):
         if env:
             nt = int(env)
         ids = None if count == n else idx.astype(np.int32)
-        out.append((nt, ids, int(batch_n[mask].max()), count))
+        out.append((nt, ids, int(batch_n[mask].max()), count, False))
     return out
+
+
+def alloc_gmem_slabs(n_blocks: int, max_n: int, nt: int, device: str):
+    """Allocate the global-memory geometry slabs for a huge-N bucket:
+    per-block chunk-transposed fp32 reciprocals + fp64 anchor log-prefixes
+    (the same layout the LDS path uses)."""
+    import torch
+
+    chunk = (max_n + nt - 1) // nt
+    ksub = (chunk + 31) // 32
+    g_inv = torch.empty((n_blocks, chunk * nt), dtype=torch.float32, device=device)
+    g_anchor = torch.empty((n_blocks, nt * ksub), dtype=torch.float64, device=device)
+    return g_inv, g_anchor
 
 _lib: Optional[ctypes.CDLL] = None
 
@@ -105,6 +137,8 @@ def load_library(allow_build: bool = True) -> ctypes.CDLL:
         ctypes.POINTER(ctypes.c_int),
         ctypes.c_int,
         ctypes.c_float,
+        ctypes.POINTER(ctypes.c_void_p),  # per-bucket g_inv slabs (huge-N)
+        ctypes.POINTER(ctypes.c_void_p),  # per-bucket g_anchor slabs
     ]
     lib.wva_reconcile.restype = ctypes.c_int
     lib.wva_reconcile.argtypes = [ctypes.c_void_p]
@@ -148,9 +182,6 @@ def run_sweep(arrays: dict, device: str = "cuda", analyzer_mode: int = 0,
         raise HipKernelError("run_sweep requires a GPU (use the CPU engine backend instead)")
     lib = load_library()
     n_cells = int(arrays["in_tok"].shape[0])
-    max_n = int(arrays["batch_n"].max().item()) if n_cells else 1
-    if max_n > MAX_N:
-        raise HipKernelError(f"batch size {max_n} exceeds kernel limit {MAX_N}")
 
     dev = {k: v.to(device, non_blocking=True) for k, v in arrays.items()}
     t = torch
@@ -168,17 +199,18 @@ def run_sweep(arrays: dict, device: str = "cuda", analyzer_mode: int = 0,
     )
     stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
 
-    # partition cells into regime-adaptive N-buckets
-    buckets = [
-        (nt, torch.from_numpy(ids).to(device) if ids is not None else None, bmax)
-        for nt, ids, bmax, _count in choose_buckets(arrays["batch_n"].numpy())
-    ]
+    # partition cells into regime-adaptive N-buckets (+ huge-N GMEM spill)
+    buckets = []
+    for nt, ids, bmax, count, gmem in choose_buckets(arrays["batch_n"].numpy()):
+        ids_t = torch.from_numpy(ids).to(device) if ids is not None else None
+        slabs = alloc_gmem_slabs(count, bmax, nt, device) if gmem else (None, None)
+        buckets.append((nt, ids_t, bmax, slabs))
 
     # overlap the bucket launches on separate HIP streams: wall time becomes
     # the straggler bucket's latency instead of the sum of all three
     main_stream = torch.cuda.current_stream()
     side = [torch.cuda.Stream() for _ in range(max(len(buckets) - 1, 0))]
-    for i, (nt, ids, bmax) in enumerate(buckets):
+    for i, (nt, ids, bmax, slabs) in enumerate(buckets):
         if i == 0:
             cur = main_stream
         else:
@@ -220,6 +252,8 @@ def run_sweep(arrays: dict, device: str = "cuda", analyzer_mode: int = 0,
             _ptr(out.ttft),
             _ptr(out.rho),
             _ptr(out.max_rate),
+            _ptr(slabs[0]) if slabs[0] is not None else None,
+            _ptr(slabs[1]) if slabs[1] is not None else None,
         )
         if rc != 0:
             raise HipKernelError(f"wva_sweep_launch_bucket(nt={nt}) failed with hipError {rc}")

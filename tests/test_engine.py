@@ -117,3 +117,59 @@ class TestCpuEngineParity:
         # empty allocation has strongly negative value (-0.9*cur cost), wins
         assert alloc.accelerator == ""
         assert alloc.num_replicas == 0
+
+
+class TestHugeBuckets:
+    """choose_buckets dispatches N > 8192 cells to the global-memory spill
+    bucket (uncapped sweep)."""
+
+    def test_huge_bucket_partition(self):
+        import numpy as np
+
+        from inferno_amd.ops.sweep import MAX_N, choose_buckets
+
+        batch_n = np.array([8, 600, 3000, 9000, 20000], dtype=np.int32)
+        buckets = choose_buckets(batch_n)
+        gmem_buckets = [b for b in buckets if b[4]]
+        lds_buckets = [b for b in buckets if not b[4]]
+        assert len(gmem_buckets) == 1
+        nt, ids, bmax, count, gmem = gmem_buckets[0]
+        assert count == 2 and bmax == 20000 and set(ids) == {3, 4}
+        assert nt in (256, 1024)
+        for _, _, bmax, _, _ in lds_buckets:
+            assert bmax <= MAX_N
+
+    def test_absurd_n_fails_loudly(self):
+        import numpy as np
+        import pytest as _pytest
+
+        from inferno_amd.ops.sweep import HUGE_MAX_N, HipKernelError, choose_buckets
+
+        with _pytest.raises(HipKernelError):
+            choose_buckets(np.array([HUGE_MAX_N + 1], dtype=np.int64))
+
+    def test_fastpath_batch_uncapped(self):
+        """FastSweep's vectorized batch sizing must agree with the scalar
+        compute_batch_size at huge N (no clamp on either path)."""
+        from inferno_amd.core.system import System
+        from inferno_amd.engine.fastpath import FastSweep
+        from inferno_amd.engine.snapshot import compute_batch_size
+        from tests.fixtures import make_spec
+
+        from inferno_amd.config import ServerLoadSpec
+
+        system, _ = System.from_spec(make_spec(n_servers=2, seed=77))
+        for srv in system.servers.values():
+            srv.load = ServerLoadSpec(arrivalRate=600.0, avgInTokens=64, avgOutTokens=3)
+        for model in system.models.values():
+            for perf in model.perf_data.values():
+                perf.maxBatchSize = 256
+                perf.atTokens = 1024
+        fs = FastSweep(system, backend="cpu")
+        arrs = fs._refresh_dynamic()
+        want = 256 * 1024 // 3
+        assert want > 8192
+        assert (arrs["batch_n"] == want).all()
+        srv = next(iter(system.servers.values()))
+        perf = system.models[srv.model_name].get_perf_data("MI355X")
+        assert compute_batch_size(srv, perf, 3) == want
