@@ -53,6 +53,11 @@
 // memory slab (GMEM=true instantiation) — the reference's N is uncapped
 // (allocation.go:80-86), so no batch size may change the computed allocation.
 #define WVA_MAX_N 8192
+// XL LDS bucket: 8192 < N <= 32768 still fits the chain geometry in CDNA4's
+// 160 KiB LDS (136.6 KiB at NT=256, N=32768) with the >64 KiB dynamic-LDS
+// opt-in (hipFuncSetAttribute); beyond that the geometry spills to global
+// memory (GMEM instantiation)
+#define WVA_XL_MAX_N 32768
 // sanity guard for the global-memory spill path (absurd profiles fail loudly
 // instead of exhausting HBM: 4M states/cell = ~16 MB geometry per cell)
 #define WVA_HUGE_MAX_N (1 << 22)
@@ -874,6 +879,25 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
 // ---------------------------------------------------------------------------
 // runtime (nt, gmem) -> template dispatch. GMEM instantiations exist for the
 // two block widths the host's huge-bucket policy uses (256, 1024).
+// one-time >64 KiB dynamic-LDS opt-in per kernel instantiation (XL bucket);
+// opts straight into the full 160 KiB so repeat launches skip the runtime call
+static int wva_optin_lds(const void *func, size_t lds) {
+  constexpr size_t kDefaultCap = 64 * 1024;
+  if (lds <= kDefaultCap) return 0;
+  static const void *done[8] = {};
+  for (int i = 0; i < 8; ++i)
+    if (done[i] == func) return 0;
+  int rc = (int)hipFuncSetAttribute(func, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                    160 * 1024);
+  if (rc != 0) return rc;
+  for (int i = 0; i < 8; ++i)
+    if (done[i] == nullptr) {
+      done[i] = func;
+      break;
+    }
+  return 0;
+}
+
 static int wva_sweep_dispatch(int n_blocks, int max_n, int nt, const int *cell_ids,
                               int analyzer_mode, float cv2, hipStream_t s,
                               const WvaCellsIn &in, const WvaCellsOut &out, float *g_inv,
@@ -881,7 +905,7 @@ static int wva_sweep_dispatch(int n_blocks, int max_n, int nt, const int *cell_i
   if (n_blocks <= 0) return 0;
   const bool gmem = g_inv != nullptr && g_anchor != nullptr;
   if (max_n < 1) return -2;
-  if (!gmem && max_n > WVA_MAX_N) return -2;
+  if (!gmem && max_n > WVA_XL_MAX_N) return -2;
   if (gmem && max_n > WVA_HUGE_MAX_N) return -4;
   const int chunk = (max_n + nt - 1) / nt;
   const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
@@ -889,9 +913,14 @@ static int wva_sweep_dispatch(int n_blocks, int max_n, int nt, const int *cell_i
   // the spill path keeps only the header in LDS
   size_t lds = gmem ? (size_t)40 * sizeof(double)
                     : (size_t)(40 + (chunk * nt + 1) / 2 + (size_t)nt * ksub) * sizeof(double);
+  if (lds > 160 * 1024) return -5;  // exceeds the CU's LDS
 #define WVA_LAUNCH(NTV, GM)                                                                 \
-  hipLaunchKernelGGL((wva_sweep_t<NTV, GM>), dim3(n_blocks), dim3(NTV), lds, s, in, out,    \
-                     n_blocks, cell_ids, max_n, analyzer_mode, cv2, g_inv, g_anchor)
+  do {                                                                                      \
+    int orc = wva_optin_lds(reinterpret_cast<const void *>(&wva_sweep_t<NTV, GM>), lds);    \
+    if (orc != 0) return orc;                                                               \
+    hipLaunchKernelGGL((wva_sweep_t<NTV, GM>), dim3(n_blocks), dim3(NTV), lds, s, in, out,  \
+                       n_blocks, cell_ids, max_n, analyzer_mode, cv2, g_inv, g_anchor);     \
+  } while (0)
   if (gmem) {
     switch (nt) {
       case 256: WVA_LAUNCH(256, true); break;
@@ -996,7 +1025,7 @@ extern "C" __global__ void __launch_bounds__(256) wva_gather(
 // three streams, argmin, gather, D2H, sync) — no per-step host dispatch
 // beyond wva_reconcile(ctx).
 // ---------------------------------------------------------------------------
-#define WVA_MAX_BUCKETS 4
+#define WVA_MAX_BUCKETS 5
 
 struct WvaBucket {
   int nt;
@@ -1029,8 +1058,10 @@ struct WvaCtx {
   void *pin_out_i;    // host pinned, 4 x n_srv i32
   WvaBucket buckets[WVA_MAX_BUCKETS];
   int n_buckets;
-  hipStream_t s0, s1, s2, s3;
-  hipEvent_t e_up, e_b1, e_b2, e_b3;
+  hipStream_t s0;                           // primary stream
+  hipStream_t side[WVA_MAX_BUCKETS - 1];    // overlap streams for buckets 1..
+  hipEvent_t e_up;                          // upload-complete
+  hipEvent_t e_b[WVA_MAX_BUCKETS - 1];      // side-bucket completion
 };
 
 // pointer-slot order for wva_ctx_create (host mirrors in ops/sweep.py):
@@ -1092,14 +1123,12 @@ extern "C" void *wva_ctx_create(int n_cells, int n_srv, void **p) {
   c->pin_f = p[27];
   c->pin_out_f = p[28];
   c->pin_out_i = p[29];
-  if (hipStreamCreateWithFlags(&c->s0, hipStreamNonBlocking) != hipSuccess ||
-      hipStreamCreateWithFlags(&c->s1, hipStreamNonBlocking) != hipSuccess ||
-      hipStreamCreateWithFlags(&c->s2, hipStreamNonBlocking) != hipSuccess ||
-      hipStreamCreateWithFlags(&c->s3, hipStreamNonBlocking) != hipSuccess ||
-      hipEventCreateWithFlags(&c->e_up, hipEventDisableTiming) != hipSuccess ||
-      hipEventCreateWithFlags(&c->e_b1, hipEventDisableTiming) != hipSuccess ||
-      hipEventCreateWithFlags(&c->e_b2, hipEventDisableTiming) != hipSuccess ||
-      hipEventCreateWithFlags(&c->e_b3, hipEventDisableTiming) != hipSuccess) {
+  bool ok = hipStreamCreateWithFlags(&c->s0, hipStreamNonBlocking) == hipSuccess &&
+            hipEventCreateWithFlags(&c->e_up, hipEventDisableTiming) == hipSuccess;
+  for (int i = 0; ok && i < WVA_MAX_BUCKETS - 1; ++i)
+    ok = hipStreamCreateWithFlags(&c->side[i], hipStreamNonBlocking) == hipSuccess &&
+         hipEventCreateWithFlags(&c->e_b[i], hipEventDisableTiming) == hipSuccess;
+  if (!ok) {
     delete c;
     return nullptr;
   }
@@ -1143,18 +1172,17 @@ extern "C" int wva_reconcile(void *ctx) {
   if (err != hipSuccess) return (int)err;
   if (hipEventRecord(c->e_up, c->s0) != hipSuccess) return -10;
 
-  // 2. bucket launches: bucket 0 on s0; others overlap on s1/s2/s3 after upload
-  hipStream_t side[3] = {c->s1, c->s2, c->s3};
-  hipEvent_t bev[3] = {c->e_b1, c->e_b2, c->e_b3};
+  // 2. bucket launches: bucket 0 on s0; others overlap on side streams after
+  // the upload event
   for (int i = 0; i < c->n_buckets; ++i) {
-    hipStream_t s = (i == 0) ? c->s0 : side[i - 1];
+    hipStream_t s = (i == 0) ? c->s0 : c->side[i - 1];
     if (i > 0 && hipStreamWaitEvent(s, c->e_up, 0) != hipSuccess) return -11;
     int rc = wva_launch_bucket_on(c, c->buckets[i], s);
     if (rc != 0) return rc;
-    if (i > 0 && hipEventRecord(bev[i - 1], s) != hipSuccess) return -12;
+    if (i > 0 && hipEventRecord(c->e_b[i - 1], s) != hipSuccess) return -12;
   }
   for (int i = 1; i < c->n_buckets; ++i)
-    if (hipStreamWaitEvent(c->s0, bev[i - 1], 0) != hipSuccess) return -14;
+    if (hipStreamWaitEvent(c->s0, c->e_b[i - 1], 0) != hipSuccess) return -14;
 
   // 3. argmin + gather on s0
   hipLaunchKernelGGL(wva_argmin, dim3(c->n_srv), dim3(WVA_WAVE), 0, c->s0, c->out.value,
@@ -1180,12 +1208,10 @@ extern "C" void wva_ctx_destroy(void *ctx) {
   WvaCtx *c = (WvaCtx *)ctx;
   if (c == nullptr) return;
   (void)hipStreamDestroy(c->s0);
-  (void)hipStreamDestroy(c->s1);
-  (void)hipStreamDestroy(c->s2);
-  (void)hipStreamDestroy(c->s3);
   (void)hipEventDestroy(c->e_up);
-  (void)hipEventDestroy(c->e_b1);
-  (void)hipEventDestroy(c->e_b2);
-  (void)hipEventDestroy(c->e_b3);
+  for (int i = 0; i < WVA_MAX_BUCKETS - 1; ++i) {
+    (void)hipStreamDestroy(c->side[i]);
+    (void)hipEventDestroy(c->e_b[i]);
+  }
   delete c;
 }

@@ -14,7 +14,10 @@ from typing import Optional
 
 from .build import LIB_PATH, build, needs_build
 
-MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip (LDS-resident limit)
+MAX_N = 8192  # must match WVA_MAX_N in wva_kernels.hip (64KB-LDS limit)
+# XL LDS tier: fits CDNA4's 160 KiB LDS with the >64KB dynamic-LDS opt-in
+# (must match WVA_XL_MAX_N); override to MAX_N to force such cells to GMEM
+XL_MAX_N = int(os.environ.get("INFERNO_XL_MAX_N", "32768"))
 HUGE_MAX_N = 1 << 22  # must match WVA_HUGE_MAX_N (global-memory spill limit)
 # N-bucket thresholds (must match WVA_N_SMALL / WVA_N_MED): cells are
 # dispatched to 64- and 256-thread blocks by batch size — widths picked by
@@ -33,9 +36,11 @@ def choose_buckets(batch_n):
     waves-per-SIMD x instructions, while the serial chain states are cheap
     fp32-LDS FMAs. Narrow blocks therefore win in BOTH regimes.
 
-    Cells with N above MAX_N (the LDS budget) go to a fourth, global-memory
-    bucket: their chain geometry lives in a per-block HBM slab instead of LDS
-    (GMEM kernel instantiation) so the sweep is uncapped like the reference
+    Cells with MAX_N < N <= XL_MAX_N use the XL tier: still LDS-resident via
+    the 160 KiB dynamic-LDS opt-in (one workgroup per CU, fine for the rare
+    huge cells). Cells above XL_MAX_N go to the global-memory bucket: their
+    chain geometry lives in a per-block HBM slab instead of LDS (GMEM kernel
+    instantiation). Either way the sweep is uncapped like the reference
     (allocation.go:80-86).
 
     Returns [(nt, cell_idx int32 array or None, bucket_max_n, count, gmem)].
@@ -47,13 +52,19 @@ def choose_buckets(batch_n):
         (64, batch_n <= N_SMALL),
         (256, (batch_n > N_SMALL) & (batch_n <= N_MED)),
         (1024, (batch_n > N_MED) & (batch_n <= MAX_N)),
-        ("huge", batch_n > MAX_N),
+        ("xl", (batch_n > MAX_N) & (batch_n <= XL_MAX_N)),
+        ("huge", batch_n > XL_MAX_N),
     ]
     out = []
     for base_nt, mask in masks:
         idx = np.nonzero(mask)[0]
         count = len(idx)
         if count == 0:
+            continue
+        if base_nt == "xl":
+            nt = int(os.environ.get("INFERNO_NT_XL", "256"))
+            ids = None if count == n else idx.astype(np.int32)
+            out.append((nt, ids, int(batch_n[mask].max()), count, False))
             continue
         if base_nt == "huge":
             bmax = int(batch_n[mask].max())

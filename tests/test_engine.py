@@ -123,21 +123,36 @@ class TestHugeBuckets:
     """choose_buckets dispatches N > 8192 cells to the global-memory spill
     bucket (uncapped sweep)."""
 
-    def test_huge_bucket_partition(self):
+    def test_bucket_partition_xl_and_gmem(self):
         import numpy as np
 
-        from inferno_amd.ops.sweep import MAX_N, choose_buckets
+        from inferno_amd.ops.sweep import MAX_N, XL_MAX_N, choose_buckets
 
-        batch_n = np.array([8, 600, 3000, 9000, 20000], dtype=np.int32)
+        batch_n = np.array([8, 600, 3000, 9000, 20000, 40000], dtype=np.int32)
         buckets = choose_buckets(batch_n)
         gmem_buckets = [b for b in buckets if b[4]]
         lds_buckets = [b for b in buckets if not b[4]]
+        # 40000 > XL_MAX_N -> global-memory spill
         assert len(gmem_buckets) == 1
         nt, ids, bmax, count, gmem = gmem_buckets[0]
+        assert count == 1 and bmax == 40000 and set(ids) == {5}
+        # 9000/20000 -> XL LDS tier (160KB dynamic-LDS opt-in)
+        xl = [b for b in lds_buckets if b[2] > MAX_N]
+        assert len(xl) == 1
+        nt, ids, bmax, count, gmem = xl[0]
         assert count == 2 and bmax == 20000 and set(ids) == {3, 4}
-        assert nt in (256, 1024)
+        assert bmax <= XL_MAX_N
         for _, _, bmax, _, _ in lds_buckets:
-            assert bmax <= MAX_N
+            assert bmax <= XL_MAX_N
+
+    def test_xl_override_forces_gmem(self, monkeypatch):
+        import numpy as np
+
+        import inferno_amd.ops.sweep as sweep
+
+        monkeypatch.setattr(sweep, "XL_MAX_N", sweep.MAX_N)
+        buckets = sweep.choose_buckets(np.array([9000], dtype=np.int32))
+        assert len(buckets) == 1 and buckets[0][4] is True
 
     def test_absurd_n_fails_loudly(self):
         import numpy as np

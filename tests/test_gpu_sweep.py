@@ -368,9 +368,10 @@ class TestConfig5ShapeDifferential:
 
 
 class TestHugeNGmemSpill:
-    """N > 8192 cells spill chain geometry to global memory (GMEM kernel
-    instantiation); results must match the CPU golden exactly like the LDS
-    path does (VERDICT r1 item 4 — the reference's N is uncapped)."""
+    """N > 8192 cells stay uncapped (VERDICT r1 item 4 — the reference's N is
+    uncapped): 8192 < N <= 32768 runs the XL tier (160KB dynamic LDS), larger
+    N spills chain geometry to global memory (GMEM kernel instantiation).
+    Both must match the CPU golden exactly like the standard LDS path."""
 
     def _set_huge(self, s, k=33):
         # N = 256*2048//33 = 15887 > 8192 on every cell
@@ -381,7 +382,8 @@ class TestHugeNGmemSpill:
         for srv in s.servers.values():
             srv.load = ServerLoadSpec(arrivalRate=240.0, avgInTokens=64, avgOutTokens=k)
 
-    def test_huge_cells_match_cpu(self):
+    def test_xl_lds_cells_match_cpu(self):
+        # default policy: 15887 -> XL LDS tier
         cpu_sys, gpu_sys, opt = build_pair(n_servers=3, seed=400)
         self._set_huge(cpu_sys)
         self._set_huge(gpu_sys)
@@ -399,9 +401,31 @@ class TestHugeNGmemSpill:
                 n_seen += 1
         assert n_seen > 0
 
-    def test_mixed_fleet_four_buckets(self):
-        """Fleet spanning all four N-buckets (64/256-LDS, 1024-LDS, GMEM) in
-        one sweep: exercises the 4-stream overlap in the native ctx too."""
+    def test_gmem_cells_match_cpu(self, monkeypatch):
+        # force the same cells down the global-memory spill path
+        import inferno_amd.ops.sweep as sweep
+
+        monkeypatch.setattr(sweep, "XL_MAX_N", sweep.MAX_N)
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=3, seed=400)
+        self._set_huge(cpu_sys)
+        self._set_huge(gpu_sys)
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        n_seen = 0
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map), f"feasibility mismatch for {name}"
+            for acc in a_map:
+                assert a_map[acc].batch_size > 8192
+                assert b_map[acc].batch_size == a_map[acc].batch_size
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+                n_seen += 1
+        assert n_seen > 0
+
+    def test_mixed_fleet_multi_bucket(self):
+        """Fleet spanning four N-buckets (small/medium/large LDS + XL) in one
+        sweep: exercises the multi-stream overlap too."""
         cpu_sys, gpu_sys, opt = build_pair(n_servers=8, seed=401)
         for s in (cpu_sys, gpu_sys):
             loads = [
@@ -431,9 +455,9 @@ class TestHugeNGmemSpill:
                 assert_alloc_close(a_map[acc], b_map[acc], name, acc)
         assert seen_huge > 0
 
-    def test_fastpath_native_ctx_with_huge_bucket(self):
-        """The persistent native reconcile ctx must handle the 4th (GMEM)
-        bucket: FastSweep GPU winners == FastSweep CPU winners."""
+    def test_fastpath_native_ctx_with_gmem_bucket(self):
+        """The persistent native reconcile ctx must handle the GMEM spill
+        bucket (N > 32768): FastSweep GPU winners == FastSweep CPU winners."""
         from inferno_amd.engine.fastpath import FastSweep
 
         cpu_sys, gpu_sys, opt = build_pair(n_servers=6, seed=402)
@@ -441,10 +465,11 @@ class TestHugeNGmemSpill:
             for i, srv_name in enumerate(sorted(s.servers)):
                 srv = s.servers[srv_name]
                 if i % 2 == 0:
+                    # N = 256*2048//9 = 58254 > XL_MAX_N -> GMEM path
                     srv.load = ServerLoadSpec(240.0, 64, 9)
                     for perf in s.models[srv.model_name].perf_data.values():
                         perf.maxBatchSize = 256
-                        perf.atTokens = 512
+                        perf.atTokens = 2048
                 else:
                     srv.load = ServerLoadSpec(60.0, 128, 300)
         rec_cpu = FastSweep(cpu_sys, backend="cpu").reconcile()
