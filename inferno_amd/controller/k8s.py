@@ -140,6 +140,10 @@ class HttpKube:
         import httpx
 
         if base_url is None:
+            # KUBE_API_URL overrides the in-cluster default (used by the
+            # envtest-equivalent e2e tier to point at the apiserver stand-in)
+            base_url = os.environ.get("KUBE_API_URL") or None
+        if base_url is None:
             host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes.default.svc")
             port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
             base_url = f"https://{host}:{port}"
