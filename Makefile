@@ -17,6 +17,12 @@ test:
 test-gpu:
 	$(PYTHON) -m pytest tests/ -x -q -m gpu
 
+## Full-loop out-of-process e2e: real controller process against the
+## kube-apiserver stand-in + vLLM emulator + TLS Prometheus stand-in
+## (envtest/Kind-equivalent tier; no cluster binaries needed)
+test-e2e:
+	$(PYTHON) -m pytest tests/test_e2e_apiserver.py tests/test_e2e_controller.py -x -q
+
 ## Flagship benchmark: 512-model fleet, GPU sweep
 bench: build
 	$(PYTHON) bench.py --steps 30 --warmup 5

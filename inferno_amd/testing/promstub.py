@@ -27,6 +27,7 @@ from typing import Optional
 _RATE_RE = re.compile(
     r"^sum\(rate\(([a-zA-Z_:][a-zA-Z0-9_:]*)\{([^}]*)\}\[(\d+)([smh])\]\)\)$"
 )
+_SELECTOR_RE = re.compile(r"^([a-zA-Z_:][a-zA-Z0-9_:]*)(?:\{([^}]*)\})?$")
 _LABEL_RE = re.compile(r'([a-zA-Z_][a-zA-Z0-9_]*)="([^"]*)"')
 
 
@@ -72,6 +73,21 @@ class SeriesDB:
                 total = (total or 0.0) + delta / (t1 - t0)
         return total
 
+    def latest(self, metric: str, matchers: dict[str, str]) -> Optional[float]:
+        """Instant-vector selector: sum of the most recent sample per
+        matching series (the collector's availability probe uses this,
+        collector.go:87-156)."""
+        total = None
+        with self._lock:
+            for (m, lset), dq in self._series.items():
+                if m != metric or not dq:
+                    continue
+                labels = dict(lset)
+                if any(labels.get(k) != v for k, v in matchers.items()):
+                    continue
+                total = (total or 0.0) + dq[-1][1]
+        return total
+
 
 def scrape_once(db: SeriesDB, target: str, client) -> None:
     from prometheus_client.parser import text_string_to_metric_families
@@ -88,23 +104,40 @@ def evaluate(db: SeriesDB, promql: str) -> Optional[float]:
     promql = promql.strip()
     if promql == "up":
         return 1.0
-    if "/" in promql and promql.count("/") == 1:
-        left, right = promql.split("/", 1)
-        a = evaluate(db, left)
-        b = evaluate(db, right)
-        if a is None or b is None or b == 0:
-            return None
-        return a / b
+    # ratio split: only a "/" at paren depth 0 outside quotes is an operator
+    # (label values like model_name="default/default" contain slashes)
+    depth = 0
+    in_q = False
+    for i, ch in enumerate(promql):
+        if ch == '"':
+            in_q = not in_q
+        elif not in_q and ch == "(":
+            depth += 1
+        elif not in_q and ch == ")":
+            depth -= 1
+        elif not in_q and ch == "/" and depth == 0:
+            a = evaluate(db, promql[:i])
+            b = evaluate(db, promql[i + 1:])
+            if a is None or b is None or b == 0:
+                return None
+            return a / b
     m = _RATE_RE.match(promql)
-    if m is None:
-        return None
-    metric, labels_s, num, unit = m.groups()
-    matchers = dict(_LABEL_RE.findall(labels_s))
+    if m is not None:
+        metric, labels_s, num, unit = m.groups()
+        matchers = _matchers(labels_s)
+        range_s = float(num) * {"s": 1, "m": 60, "h": 3600}[unit]
+        return db.sum_rate(metric, matchers, range_s)
+    m = _SELECTOR_RE.match(promql)
+    if m is not None:
+        metric, labels_s = m.groups()
+        return db.latest(metric, _matchers(labels_s or ""))
+    return None
+
+
+def _matchers(labels_s: str) -> dict[str, str]:
     # drop empty-string matchers (the collector emits namespace="" when the
     # emulator exports no namespace label)
-    matchers = {k: v for k, v in matchers.items() if v != ""}
-    range_s = float(num) * {"s": 1, "m": 60, "h": 3600}[unit]
-    return db.sum_rate(metric, matchers, range_s)
+    return {k: v for k, v in _LABEL_RE.findall(labels_s) if v != ""}
 
 
 def build_app(db: SeriesDB):

@@ -8,6 +8,11 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: test requires an AMD MI355X GPU (run via gpurun)")
+    config.addinivalue_line(
+        "markers",
+        "e2e: full-loop out-of-process e2e (controller + apiserver stand-in + "
+        "emulator + prometheus stand-in); CPU-only but slower",
+    )
 
 
 def pytest_collection_modifyitems(config, items):

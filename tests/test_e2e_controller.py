@@ -1,0 +1,319 @@
+"""Full-loop e2e: the REAL controller process against the apiserver stand-in,
+the vLLM emulator and a TLS Prometheus stand-in, under real HTTP load.
+
+The executed counterpart of the reference's Kind e2e
+(test/e2e/e2e_test.go:341-430 scale-out under load with Prometheus
+cross-checks, :519 scale-in at idle, conditions/ownerRef/lease assertions)
+— every component in its own process, wire formats end to end.
+"""
+from __future__ import annotations
+
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import threading
+import time
+
+import httpx
+import pytest
+import yaml
+
+from tests.test_e2e_apiserver import CRD, NS_SYS, REPO, _apply, _free_port, _spawn
+
+VA_NS = "llm-d-sim"
+VA_NAME = "vllme-deploy"
+MODEL = "default/default"
+
+
+def _mk_cert(tmp_path):
+    import subprocess as sp
+
+    key = tmp_path / "tls.key"
+    crt = tmp_path / "tls.crt"
+    sp.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(crt), "-days", "1",
+         "-subj", "/CN=promstub",
+         "-addext", "subjectAltName=DNS:localhost,IP:127.0.0.1"],
+        check=True, capture_output=True,
+    )
+    return str(crt), str(key)
+
+
+@pytest.fixture(scope="module")
+def world(tmp_path_factory):
+    """apiserver + emulator + TLS promstub + seeded objects + controller."""
+    tmp = tmp_path_factory.mktemp("e2e")
+    procs = []
+    try:
+        # 1. apiserver stand-in
+        api_proc, api_port = _spawn(
+            [sys.executable, "-m", "inferno_amd.testing.kubeapi", "--port", "0",
+             "--crd", CRD],
+            match="kubeapi listening on",
+        )
+        procs.append(api_proc)
+        kube = httpx.Client(base_url=f"http://127.0.0.1:{api_port}", timeout=10.0)
+
+        # 2. vLLM emulator (fast decode so load generation is cheap)
+        emu_port = _free_port()
+        emu_env = {
+            "PORT": str(emu_port), "MODEL_NAME": MODEL, "NAMESPACE": VA_NS,
+            "DECODE_TIME": "1", "PREFILL_TIME": "1", "MAX_BATCH_SIZE": "256",
+        }
+        emu_proc = subprocess.Popen(
+            [sys.executable, "-m", "inferno_amd.emulator.server"],
+            env={**os.environ, **emu_env},
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, cwd=REPO,
+        )
+        procs.append(emu_proc)
+        emu = f"http://127.0.0.1:{emu_port}"
+        for _ in range(100):
+            try:
+                if httpx.get(f"{emu}/healthz", timeout=2).status_code == 200:
+                    break
+            except httpx.HTTPError:
+                time.sleep(0.2)
+        else:
+            raise RuntimeError("emulator did not come up")
+
+        # 3. TLS promstub scraping the emulator (reference enforces HTTPS
+        #    Prometheus, tls.go:63-68 — keep that property end to end)
+        crt, key = _mk_cert(tmp)
+        prom_proc, prom_port = _spawn(
+            [sys.executable, "-m", "inferno_amd.testing.promstub", "--port", "0",
+             "--target", emu, "--interval", "0.5",
+             "--tls-cert", crt, "--tls-key", key],
+            match="promstub listening on",
+        )
+        procs.append(prom_proc)
+
+        # 4. seed cluster objects
+        for doc in yaml.safe_load_all(
+            open(os.path.join(REPO, "deploy", "configmap-accelerator-unitcost.yaml"))
+        ):
+            if doc:
+                assert _apply(kube, doc).status_code == 201
+        for doc in yaml.safe_load_all(
+            open(os.path.join(REPO, "deploy", "configmap-serviceclass.yaml"))
+        ):
+            if doc:
+                assert _apply(kube, doc).status_code == 201
+        assert _apply(kube, {
+            "apiVersion": "v1", "kind": "ConfigMap",
+            "metadata": {
+                "name": "workload-variant-autoscaler-variantautoscaling-config",
+                "namespace": NS_SYS,
+            },
+            "data": {"GLOBAL_OPT_INTERVAL": "2s"},
+        }).status_code == 201
+        dep = _apply(kube, {
+            "apiVersion": "apps/v1", "kind": "Deployment",
+            "metadata": {"name": VA_NAME, "namespace": VA_NS},
+            "spec": {"replicas": 1}, "status": {"replicas": 1},
+        })
+        assert dep.status_code == 201
+        with open(os.path.join(REPO, "deploy", "examples",
+                               "vllme-variantautoscaling.yaml")) as f:
+            va_doc = next(d for d in yaml.safe_load_all(f)
+                          if d and d["metadata"]["name"] == VA_NAME)
+        va_doc["metadata"]["namespace"] = VA_NS
+        assert _apply(kube, va_doc).status_code == 201
+
+        # 5. the controller process
+        metrics_port = _free_port()
+        ctl_env = {
+            "KUBE_API_URL": f"http://127.0.0.1:{api_port}",
+            "PROMETHEUS_BASE_URL": f"https://127.0.0.1:{prom_port}",
+            "PROMETHEUS_CA_CERT_PATH": crt,
+            "LOG_LEVEL": "info",
+        }
+        ctl_proc = subprocess.Popen(
+            [sys.executable, "-m", "inferno_amd.controller.main",
+             "--metrics-port", str(metrics_port), "--backend", "cpu"],
+            env={**os.environ, **ctl_env},
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True, cwd=REPO,
+        )
+        procs.append(ctl_proc)
+        threading.Thread(target=lambda: [None for _ in ctl_proc.stdout],
+                         daemon=True).start()
+        # readiness
+        for _ in range(150):
+            try:
+                if httpx.get(f"http://127.0.0.1:{metrics_port}/readyz",
+                             timeout=2).status_code == 200:
+                    break
+            except httpx.HTTPError:
+                pass
+            if ctl_proc.poll() is not None:
+                raise RuntimeError("controller exited early")
+            time.sleep(0.2)
+        else:
+            raise RuntimeError("controller never became ready")
+
+        yield {
+            "kube": kube, "emu": emu,
+            "metrics_port": metrics_port, "api_port": api_port,
+        }
+        kube.close()
+    finally:
+        for p in procs:
+            try:
+                p.send_signal(signal.SIGTERM)
+            except OSError:
+                pass
+        for p in procs:
+            try:
+                p.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                p.kill()
+
+
+def _va_status(kube) -> dict:
+    r = kube.get(
+        f"/apis/llmd.ai/v1alpha1/namespaces/{VA_NS}/variantautoscalings/{VA_NAME}"
+    )
+    assert r.status_code == 200, r.text
+    return r.json()
+
+
+def _drive_load(emu: str, seconds: float, concurrency: int = 8) -> int:
+    """Hammer the emulator's OpenAI endpoint from several threads."""
+    stop = time.time() + seconds
+    done = [0]
+    lock = threading.Lock()
+
+    def worker():
+        with httpx.Client(timeout=30.0) as c:
+            while time.time() < stop:
+                try:
+                    c.post(f"{emu}/v1/chat/completions", json={
+                        "model": MODEL,
+                        "messages": [{"role": "user", "content": "lorem " * 100}],
+                        "max_tokens": 40,
+                    })
+                    with lock:
+                        done[0] += 1
+                except httpx.HTTPError:
+                    pass
+
+    threads = [threading.Thread(target=worker) for _ in range(concurrency)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    return done[0]
+
+
+def _wait_for(pred, timeout_s: float, interval: float = 1.0, desc: str = ""):
+    deadline = time.time() + timeout_s
+    last = None
+    while time.time() < deadline:
+        last = pred()
+        if last:
+            return last
+        time.sleep(interval)
+    raise AssertionError(f"timed out waiting for {desc}; last={last!r}")
+
+
+@pytest.mark.e2e
+class TestControllerEndToEnd:
+    def test_full_loop(self, world):
+        kube = world["kube"]
+        emu = world["emu"]
+
+        # -- leader lease appears (e2e_suite_test.go:95-110) -------------
+        def lease_held():
+            r = kube.get(
+                f"/apis/coordination.k8s.io/v1/namespaces/{NS_SYS}"
+                "/leases/72dd1cf1.llm-d.ai"
+            )
+            if r.status_code != 200:
+                return None
+            return r.json()["spec"].get("holderIdentity") or None
+
+        holder = _wait_for(lease_held, 30, desc="leader lease")
+        assert holder
+
+        # -- drive load; controller reconciles every 2s ------------------
+        n = _drive_load(emu, seconds=12.0, concurrency=8)
+        assert n > 20, f"load generator only completed {n} requests"
+
+        # -- scale-out: desired allocation computed and written ----------
+        def optimized():
+            va = _va_status(kube)
+            st = va.get("status", {})
+            des = st.get("desiredOptimizedAlloc", {})
+            conds = {c["type"]: c["status"] for c in st.get("conditions", [])}
+            if (des.get("numReplicas", 0) >= 1
+                    and conds.get("OptimizationReady") == "True"
+                    and conds.get("MetricsAvailable") == "True"):
+                return va
+            return None
+
+        va = _wait_for(optimized, 45, desc="desired allocation in VA status")
+        st = va["status"]
+        assert st["desiredOptimizedAlloc"]["accelerator"] == "A100"
+        assert st["actuation"]["applied"] is True
+        # collected load made it into currentAlloc (2-decimal strings)
+        assert float(st["currentAlloc"]["load"]["arrivalRate"]) > 0
+        assert st["currentAlloc"]["maxBatch"] == 256
+        # ownerReference set from the deployment (controller.go:278-293)
+        refs = va["metadata"].get("ownerReferences", [])
+        assert refs and refs[0]["kind"] == "Deployment" and refs[0]["name"] == VA_NAME
+
+        # -- controller /metrics exposes the inferno_* gauges ------------
+        body = httpx.get(
+            f"http://127.0.0.1:{world['metrics_port']}/metrics", timeout=5
+        ).text
+        assert "inferno_desired_replicas" in body
+        assert f'variant_name="{VA_NAME}"' in body
+        des_line = next(
+            line for line in body.splitlines()
+            if line.startswith("inferno_desired_replicas") and VA_NAME in line
+        )
+        assert float(des_line.rsplit(" ", 1)[-1]) >= 1.0
+
+        # -- scale-in at idle (e2e_test.go:519): the [1m] rate window ----
+        # empties after load stops; desired returns to the 1-replica floor
+        def scaled_in():
+            va = _va_status(kube)
+            des = va["status"]["desiredOptimizedAlloc"]
+            rate = float(va["status"]["currentAlloc"]["load"]["arrivalRate"] or 0)
+            return va if (des["numReplicas"] <= 1 and rate < 0.5) else None
+
+        _wait_for(scaled_in, 120, interval=2.0, desc="scale-in at idle")
+
+    def test_config5_multi_va(self, world):
+        """Second VA joins the fleet mid-flight (multi-VA scenario,
+        e2e_test.go:698-1130 shape): both get optimized statuses."""
+        kube = world["kube"]
+        # second deployment+VA pointing at the same emulator metrics
+        assert _apply(kube, {
+            "apiVersion": "apps/v1", "kind": "Deployment",
+            "metadata": {"name": "vllme-deploy-b", "namespace": VA_NS},
+            "spec": {"replicas": 1}, "status": {"replicas": 1},
+        }).status_code == 201
+        with open(os.path.join(REPO, "deploy", "examples",
+                               "vllme-variantautoscaling.yaml")) as f:
+            va_doc = next(d for d in yaml.safe_load_all(f)
+                          if d and d["metadata"]["name"] == VA_NAME)
+        va_doc["metadata"]["name"] = "vllme-deploy-b"
+        va_doc["metadata"]["namespace"] = VA_NS
+        assert _apply(kube, va_doc).status_code == 201
+
+        def second_optimized():
+            r = kube.get(
+                f"/apis/llmd.ai/v1alpha1/namespaces/{VA_NS}"
+                "/variantautoscalings/vllme-deploy-b"
+            )
+            if r.status_code != 200:
+                return None
+            st = r.json().get("status", {})
+            conds = {c["type"]: c["status"] for c in st.get("conditions", [])}
+            return r.json() if conds.get("OptimizationReady") == "True" else None
+
+        _wait_for(second_optimized, 45, desc="second VA optimized")
