@@ -133,6 +133,10 @@ class Reconciler:
         self._requested_backend = backend
         self._degraded_ticks = 0
         self.gpu_reprobe_interval = 10
+        # thread-pool width for the I/O-bound per-VA collection phase
+        import os as _os
+
+        self.collect_workers = int(_os.environ.get("WVA_COLLECT_WORKERS", "8"))
 
     # ------------------------------------------------------------------
     def read_interval(self) -> float:
@@ -216,29 +220,24 @@ class Reconciler:
                 continue
 
         # ---- prepare phase (per-VA, continue on error) -------------------
-        update_list: list[api.VariantAutoscaling] = []
-        for va in vas:
+        # Collection is I/O-bound (5 PromQL + k8s gets per VA, ref
+        # controller.go:218-335 does them serially); here the per-VA network
+        # phase fans out over a thread pool and only the shared-spec mutation
+        # runs serially. Per-VA failures never block the fleet.
+        def _collect(va: api.VariantAutoscaling):
             model_name = va.spec.modelID
             if not model_name:
-                continue
+                return None
             try:
                 _, class_name = adapters.find_model_slo(service_class_cm, model_name)
             except adapters.AdapterError as e:
-                result.errors.append(f"{va.name}: {e}")
-                continue
-            for profile in va.spec.modelProfile.accelerators:
-                try:
-                    adapters.add_model_accelerator_profile(spec, model_name, profile)
-                except adapters.AdapterError:
-                    continue
+                return f"{va.name}: {e}"
             acc_name = va.labels.get(api.ACCELERATOR_LABEL, "")
             if acc_name not in acc_costs:
-                result.errors.append(f"{va.name}: missing accelerator cost for {acc_name!r}")
-                continue
+                return f"{va.name}: missing accelerator cost for {acc_name!r}"
             deploy = self.kube.get_deployment(va.namespace, va.name)
             if deploy is None:
-                result.errors.append(f"{va.name}: deployment not found")
-                continue
+                return f"{va.name}: deployment not found"
             if not any(r.get("uid") == deploy.uid for r in va.ownerReferences):
                 self.kube.set_owner_reference(va, deploy)
 
@@ -255,18 +254,39 @@ class Reconciler:
                 )
             else:
                 # metrics unavailable: log and skip (ref controller.go:305-316)
-                result.errors.append(f"{va.name}: metrics unavailable ({validation.reason})")
-                continue
+                return f"{va.name}: metrics unavailable ({validation.reason})"
 
             try:
                 current_alloc = collector.add_metrics_to_opt_status(
                     va, deploy.namespace, deploy.replicas, acc_costs[acc_name], self.prom
                 )
             except Exception as e:
-                result.errors.append(f"{va.name}: metric collection failed: {e}")
-                continue
+                return f"{va.name}: metric collection failed: {e}"
             va.status.currentAlloc = current_alloc
+            return (va, class_name)
 
+        if self.collect_workers > 1 and len(vas) > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            with ThreadPoolExecutor(max_workers=self.collect_workers) as pool:
+                collected = list(pool.map(_collect, vas))
+        else:
+            collected = [_collect(va) for va in vas]
+
+        update_list: list[api.VariantAutoscaling] = []
+        for va, item in zip(vas, collected):
+            if item is None:
+                continue
+            if isinstance(item, str):
+                result.errors.append(item)
+                continue
+            va, class_name = item
+            # serial: these mutate the shared SystemSpec
+            for profile in va.spec.modelProfile.accelerators:
+                try:
+                    adapters.add_model_accelerator_profile(spec, va.spec.modelID, profile)
+                except adapters.AdapterError:
+                    continue
             try:
                 adapters.add_server_info(spec, va, class_name, self.scale_to_zero)
             except adapters.AdapterError as e:

@@ -89,13 +89,63 @@ def prom_for(n_vas, rates, rng_tok):
     return MockPromAPI(results=results)
 
 
+class LatencyProm:
+    """MockPromAPI wrapper simulating network latency per query (the
+    full-reconcile-loop bench: 5 PromQL + availability probe per VA,
+    ref controller.go:86-201)."""
+
+    def __init__(self, inner, latency_ms: float):
+        self.inner = inner
+        self.latency_s = latency_ms / 1000.0
+
+    def query(self, promql):
+        if self.latency_s > 0:
+            time.sleep(self.latency_s)
+        return self.inner.query(promql)
+
+
+def bench_full_loop(args, rng):
+    """--bench mode: time the FULL reconcile loop —
+    prepare -> collect (mocked-latency PromQL) -> solve -> status write —
+    at fleet scale on the requested backend; report p50/p95."""
+    kube = make_world(args.vas, np.random.default_rng(args.seed))
+    rec = Reconciler(kube, None, MetricsEmitter(registry=CollectorRegistry()),
+                     backend=args.backend, scale_to_zero=False)
+    tok = np.random.default_rng(args.seed + 1).integers(8, 2048, size=(args.vas, 2))
+    lat = []
+    for t in range(args.ticks):
+        rates = rng.gamma(1.5, 2.0, args.vas) * rng.choice(
+            [0.0, 0.2, 1.0, 5.0], args.vas, p=[0.1, 0.3, 0.4, 0.2])
+        rec.prom = LatencyProm(prom_for(args.vas, rates, tok), args.prom_latency_ms)
+        t0 = time.perf_counter()
+        res = rec.reconcile()
+        lat.append((time.perf_counter() - t0) * 1000)
+        if res.errors[:1] and t == 0:
+            print("tick errors:", res.errors[:3])
+    lat.sort()
+    p50 = lat[len(lat) // 2]
+    p95 = lat[max(int(len(lat) * 0.95) - 1, 0)]
+    print(f"full-loop bench: vas={args.vas} ticks={args.ticks} "
+          f"backend={rec.engine.backend} prom_latency_ms={args.prom_latency_ms} "
+          f"collect_workers={rec.collect_workers} "
+          f"p50={p50:.2f}ms p95={p95:.2f}ms")
+    return 0
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--vas", type=int, default=64)
     p.add_argument("--ticks", type=int, default=200)
     p.add_argument("--seed", type=int, default=20260913)
+    p.add_argument("--bench", action="store_true",
+                   help="time the full reconcile loop only (no differential)")
+    p.add_argument("--backend", choices=["auto", "gpu", "cpu"], default="auto")
+    p.add_argument("--prom-latency-ms", type=float, default=0.0,
+                   help="simulated per-PromQL-query latency")
     args = p.parse_args()
     rng = np.random.default_rng(args.seed)
+    if args.bench:
+        return bench_full_loop(args, rng)
 
     worlds = {}
     recs = {}
