@@ -86,9 +86,17 @@ class VLLMSim:
         kv_mb_per_token: float = DEFAULT_KV_MB_PER_TOKEN,
         usable_ratio: float = DEFAULT_USABLE_MEM_RATIO,
         max_batch_size: int = DEFAULT_MAX_BATCH_SIZE,
+        decode_parms: Optional[tuple[float, float]] = None,
+        prefill_parms: Optional[tuple[float, float]] = None,
     ):
         self.decode_time_s = decode_time_ms / 1000.0
         self.prefill_time_s = prefill_time_ms / 1000.0
+        # optional batch-dependent service model (the WVA perf equations:
+        # decode(b) = alpha + beta*b, prefill(b) = gamma + delta*inTok*b, ms)
+        # — lets the emulator act as an oracle for parameter estimation
+        # (ref docs/tutorials/parameter-estimation.md:80-195)
+        self.decode_parms = decode_parms  # (alpha_ms, beta_ms)
+        self.prefill_parms = prefill_parms  # (gamma_ms, delta_ms)
         self.device = Device(mem_size_mb, kv_mb_per_token, usable_ratio)
         self.max_batch_size = max_batch_size
         self.clock = 0.0
@@ -157,10 +165,20 @@ class VLLMSim:
         memory pressure. Advances the clock by the iteration time."""
         self._admit()
 
-        step_time = self.decode_time_s
+        b = len(self.running)
+        if self.decode_parms is not None and b > 0:
+            alpha, beta = self.decode_parms
+            step_time = (alpha + beta * b) / 1000.0
+        else:
+            step_time = self.decode_time_s
         new = [r for r in self.running if not r.prefilled]
         if new:
-            step_time += self.prefill_time_s
+            if self.prefill_parms is not None:
+                gamma, delta = self.prefill_parms
+                avg_in = sum(r.input_tokens for r in new) / len(new)
+                step_time += (gamma + delta * avg_in * len(new)) / 1000.0
+            else:
+                step_time += self.prefill_time_s
             for r in new:
                 r.prefilled = True
 

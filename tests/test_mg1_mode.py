@@ -134,3 +134,67 @@ class TestDegenerateParms:
         cfg = Configuration(64, 640, ServiceParms(PrefillParms(1.0, 0.001), DecodeParms(2.0, -0.5)))
         with _pt.raises(AnalyzerError):
             QueueAnalyzer(cfg, RequestSize(10, 10))
+
+
+class TestTextbookVectors:
+    """External oracle for the M/G/1/K evaluator:
+    classical finite-capacity queueing results (Gross & Harris, 'Fundamentals
+    of Queueing Theory', M/M/1/K section; Pollaczek-Khinchine mean-value
+    formula), computed analytically with exact fractions — independent of the
+    implementation's own algebra."""
+
+    def test_mm1k_gross_harris_vector(self):
+        # M/M/1/3, lambda=3, mu=4 (rho=3/4):
+        #   p0 = 64/175, pK = 27/175, L = 201/175,
+        #   X = 444/175, W = 201/444, Wq = 15/74
+        from inferno_amd.analyzer.mm1k import MM1K
+
+        st = MM1K(3).solve(3.0, 4.0)
+        assert st.is_valid
+        assert st.p0 == pytest.approx(64 / 175, rel=1e-12)
+        assert st.pK == pytest.approx(27 / 175, rel=1e-12)
+        assert st.avg_num_in_system == pytest.approx(201 / 175, rel=1e-9)
+        assert st.throughput == pytest.approx(444 / 175, rel=1e-12)
+        assert st.avg_resp_time == pytest.approx(201 / 444, rel=1e-9)
+        assert st.avg_wait_time == pytest.approx(15 / 74, rel=1e-9)
+
+    def test_mm11_erlang_loss(self):
+        # M/M/1/1 (pure loss system) at rho=1/2: p0=2/3, p1=1/3,
+        # X = lambda*(1-p1) = 1/3, L = 1/3, W = 1/mu = 1, Wq = 0.
+        # (rho=1 with K=1 is OUTSIDE the reference's validity region
+        # rho < rhoMax=K, queuemodel.go:27-37 — also asserted here.)
+        from inferno_amd.analyzer.mm1k import MM1K
+
+        st = MM1K(1).solve(0.5, 1.0)
+        assert st.is_valid
+        assert st.p0 == pytest.approx(2 / 3, rel=1e-12)
+        assert st.pK == pytest.approx(1 / 3, rel=1e-12)
+        assert st.throughput == pytest.approx(1 / 3, rel=1e-12)
+        assert st.avg_num_in_system == pytest.approx(1 / 3, rel=1e-9)
+        assert st.avg_resp_time == pytest.approx(1.0, rel=1e-9)
+        assert st.avg_wait_time == pytest.approx(0.0, abs=1e-12)
+        assert MM1K(1).solve(1.0, 1.0).is_valid is False
+
+    def test_pollaczek_khinchine_limit(self):
+        # Large K -> unbounded M/G/1; PK mean wait Wq = rho/(1-rho) *
+        # (1+cv2)/2 * (1/mu). rho=1/2, mu=1:
+        #   cv2=1 (exponential): Wq = 1.0 (M/M/1)
+        #   cv2=0 (deterministic): Wq = 0.5
+        #   cv2=4 (heavy-tailed):  Wq = 2.5
+        from inferno_amd.analyzer.mm1k import MG1K
+
+        for cv2, want in ((1.0, 1.0), (0.0, 0.5), (4.0, 2.5)):
+            st = MG1K(10000, cv2=cv2).solve(0.5, 1.0)
+            assert st.is_valid
+            assert st.avg_wait_time == pytest.approx(want, rel=1e-6), f"cv2={cv2}"
+
+    def test_mm1k_utilization_one(self):
+        # rho == 1 degenerate case: p_i = 1/(K+1) uniformly; K=4:
+        # p0=pK=1/5, X = lambda*(1-pK) = 4/5, L = K/2 = 2
+        from inferno_amd.analyzer.mm1k import MM1K
+
+        st = MM1K(4).solve(1.0, 1.0)
+        assert st.p0 == pytest.approx(0.2, rel=1e-9)
+        assert st.pK == pytest.approx(0.2, rel=1e-9)
+        assert st.throughput == pytest.approx(0.8, rel=1e-9)
+        assert st.avg_num_in_system == pytest.approx(2.0, rel=1e-9)
