@@ -13,6 +13,9 @@ import sys
 
 OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 HIP_SRC = os.path.join(OPS_DIR, "hip", "wva_kernels.hip")
+# host-only native sources linked into the same .so (C++ runtime pieces:
+# the sequential greedy tail of limited mode)
+NATIVE_SRCS = [os.path.join(OPS_DIR, "native", "greedy.cpp")]
 LIB_DIR = os.path.join(OPS_DIR, "lib")
 LIB_PATH = os.path.join(LIB_DIR, "libwva_hip.so")
 
@@ -23,7 +26,10 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 def needs_build() -> bool:
     if not os.path.exists(LIB_PATH):
         return True
-    return os.path.getmtime(HIP_SRC) > os.path.getmtime(LIB_PATH)
+    lib_mtime = os.path.getmtime(LIB_PATH)
+    return any(
+        os.path.getmtime(src) > lib_mtime for src in [HIP_SRC, *NATIVE_SRCS]
+    )
 
 
 def build(verbose: bool = True, force: bool = False) -> str:
@@ -42,6 +48,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
         # (block max init at -inf) and exact within-tolerance comparisons.
         "-fno-gpu-rdc",
         HIP_SRC,
+        *NATIVE_SRCS,
         "-o",
         LIB_PATH,
     ]

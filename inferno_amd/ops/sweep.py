@@ -153,6 +153,7 @@ def load_library(allow_build: bool = True) -> ctypes.CDLL:
     ]
     lib.wva_reconcile.restype = ctypes.c_int
     lib.wva_reconcile.argtypes = [ctypes.c_void_p]
+    lib.wva_greedy_solve.restype = ctypes.c_int
     lib.wva_ctx_destroy.restype = None
     lib.wva_ctx_destroy.argtypes = [ctypes.c_void_p]
     _lib = lib
@@ -293,6 +294,37 @@ def run_argmin(value, feasible, seg_start) -> "object":
     if rc != 0:
         raise HipKernelError(f"wva_argmin_launch failed with hipError {rc}")
     return winner
+
+
+def run_greedy_native(cand_value, cand_acc_type, cand_units, cand_replicas,
+                      seg_start, srv_priority, capacity, delayed: bool,
+                      policy: int, allow_build: bool = True):
+    """Host-side native greedy solver (C++ in the same .so; no GPU needed).
+
+    All array args are contiguous numpy arrays (value f32, the rest i32);
+    ``capacity`` is mutated in place. Returns (winner_cand, winner_replicas)
+    int32 arrays per server (-1 = unallocated). Semantics identical to
+    solver.greedy.solve_greedy (differential-tested)."""
+    import numpy as np
+
+    lib = load_library(allow_build=allow_build)
+    n_servers = len(seg_start) - 1
+    out_cand = np.full(n_servers, -1, dtype=np.int32)
+    out_reps = np.zeros(n_servers, dtype=np.int32)
+
+    def p(a):
+        return a.ctypes.data_as(ctypes.c_void_p)
+
+    rc = lib.wva_greedy_solve(
+        ctypes.c_int(n_servers), ctypes.c_int(len(capacity)),
+        p(cand_value), p(cand_acc_type), p(cand_units), p(cand_replicas),
+        p(seg_start), p(srv_priority), p(capacity),
+        ctypes.c_int(1 if delayed else 0), ctypes.c_int(int(policy)),
+        p(out_cand), p(out_reps),
+    )
+    if rc != 0:
+        raise HipKernelError(f"wva_greedy_solve failed: {rc}")
+    return out_cand, out_reps
 
 
 def library_loaded() -> bool:

@@ -308,7 +308,9 @@ class ShardedSolver:
     def _solve_slow(self, system: System, local_names, spec: OptimizerSpec, acc_names):
         # greedy limited mode on GPU: candidates from the cached FastSweep
         # (K4 plan: GPU sweep produces the sorted-candidate inputs, the
-        # sequential capacity loop stays host-side)
+        # sequential capacity loop stays host-side — in native C++
+        # (ops/native/greedy.cpp) at world 1, Python-object greedy with
+        # all-gathered candidates at world > 1)
         if self.engine.backend == "gpu" and not spec.unlimited:
             key = (id(system), tuple(local_names))
             if self._fast_key != key:
@@ -318,13 +320,140 @@ class ShardedSolver:
                 self._fast_key = key
             cells = self._fast_sweep.reconcile_cells()
             if cells is not None:
-                self._populate_from_cells(system, local_names, cells)
                 stats = _ShardStats(
                     n_cells=self._fast_sweep.n_cells, n_servers=len(local_names)
                 )
+                initialized, _rank, world = self._dist_info()
+                import os
+
+                if (world == 1
+                        and os.environ.get("INFERNO_NATIVE_GREEDY", "1") != "0"):
+                    rec = self._solve_limited_native(system, local_names, cells,
+                                                     spec, acc_names)
+                    if rec is not None:
+                        return rec, stats
+                self._populate_from_cells(system, local_names, cells)
                 return self._finish_slow(system, local_names, spec, acc_names, stats)
         stats = self.engine.sweep(system, server_names=local_names)
         return self._finish_slow(system, local_names, spec, acc_names, stats)
+
+    def _greedy_static(self, system: System):
+        """Static per-cell greedy inputs (units/replica, accelerator-type
+        index) and per-server priorities, cached per fleet topology."""
+        fs = self._fast_sweep
+        cached = getattr(fs, "_greedy_static_cache", None)
+        if cached is not None:
+            return cached
+        type_names = sorted(
+            {acc.type for acc in system.accelerators.values()} | set(system.capacity)
+        )
+        type_index = {t: i for i, t in enumerate(type_names)}
+        acc_tidx = np.empty(len(fs.acc_names), dtype=np.int32)
+        for i, an in enumerate(fs.acc_names):
+            acc_tidx[i] = type_index[system.accelerators[an].type]
+        units = np.empty(fs.n_cells, dtype=np.int32)
+        for k in range(fs.n_cells):
+            srv = fs._srv_objs[fs.cell_server[k]]
+            acc_name = fs.acc_names[fs.cell_acc_idx[k]]
+            acc = system.accelerators[acc_name]
+            model = system.models[srv.model_name]
+            units[k] = model.get_num_instances(acc_name) * acc.multiplicity
+        prio = np.array([s.priority(system) for s in fs._srv_objs], dtype=np.int32)
+        cell_tidx = acc_tidx[fs.cell_acc_idx]
+        cached = (type_names, cell_tidx, units, prio)
+        fs._greedy_static_cache = cached
+        return cached
+
+    def _solve_limited_native(self, system: System, local_names, cells, spec,
+                              acc_names):
+        """Native limited-mode solve: vectorized candidate prep + the C++
+        greedy (ops/native/greedy.cpp) + winner-only materialization.
+        Returns a WinnerRecord, or None if the native library is missing."""
+        from ..config import SaturationPolicy
+        from ..core import Allocation
+
+        try:
+            from ..ops.sweep import run_greedy_native
+        except Exception:
+            return None
+
+        type_names, cell_tidx, units, prio = self._greedy_static(system)
+        fs = self._fast_sweep
+
+        feas = cells["feasible"].astype(bool)
+        idx = np.nonzero(feas)[0]
+        value = cells["value"]
+        srv_of = cells["cell_server"]
+        # per-server candidates sorted by value; stable -> acc-name tie order
+        # (identical to sorted(all_allocations.values(), key=value))
+        order = idx[np.lexsort((value[idx].astype(np.float64), srv_of[idx]))]
+        n_srv = len(local_names)
+        counts = np.bincount(srv_of[order], minlength=n_srv)
+        seg = np.zeros(n_srv + 1, dtype=np.int32)
+        np.cumsum(counts, out=seg[1:])
+
+        cand_value = np.ascontiguousarray(value[order], dtype=np.float32)
+        # zero-load "" allocations drop like the Python acc-lookup miss
+        cand_tidx = np.where(
+            cells["zero_empty"][order].astype(bool), -1, cell_tidx[order]
+        ).astype(np.int32)
+        cand_units = np.ascontiguousarray(units[order], dtype=np.int32)
+        cand_reps = np.ascontiguousarray(cells["num_replicas"][order], dtype=np.int32)
+        capacity = np.array(
+            [int(system.capacity.get(t, 0)) for t in type_names], dtype=np.int32
+        )
+        policy = {
+            SaturationPolicy.NONE: 0,
+            SaturationPolicy.PRIORITY_EXHAUSTIVE: 1,
+            SaturationPolicy.PRIORITY_ROUND_ROBIN: 2,
+            SaturationPolicy.ROUND_ROBIN: 3,
+        }[SaturationPolicy.parse(spec.saturationPolicy)]
+
+        try:
+            win_cand, win_reps = run_greedy_native(
+                cand_value, cand_tidx, cand_units, cand_reps, seg,
+                np.ascontiguousarray(prio, dtype=np.int32), capacity,
+                bool(spec.delayedBestEffort), policy, allow_build=False,
+            )
+        except Exception:
+            return None
+
+        # winner-only materialization (<= one Allocation per server)
+        rec = _empty_winner(n_srv)
+        acc_index = {n: i for i, n in enumerate(acc_names)}
+        servers = system.servers
+        for s in range(n_srv):
+            servers[local_names[s]].remove_allocation()
+        picked = np.nonzero(win_cand >= 0)[0]
+        for s in picked.tolist():
+            cell = int(order[win_cand[s]])
+            acc_key = fs.acc_names[int(fs.cell_acc_idx[cell])]
+            zero = bool(cells["zero_empty"][cell])
+            alloc = Allocation(
+                accelerator="" if zero else acc_key,
+                num_replicas=int(cells["num_replicas"][cell]),
+                batch_size=int(cells["batch"][cell]),
+                cost=float(cells["cost"][cell]),
+                value=float(cells["value"][cell]),
+                itl=float(cells["itl"][cell]),
+                ttft=float(cells["ttft"][cell]),
+                rho=float(cells["rho"][cell]),
+                max_arrv_rate_per_replica=float(cells["max_rate"][cell]),
+            )
+            granted = int(win_reps[s])
+            if alloc.num_replicas > 0 and granted != alloc.num_replicas:
+                factor = float(granted) / float(alloc.num_replicas)
+                alloc.cost *= factor
+                alloc.value *= factor
+                alloc.num_replicas = granted
+            servers[local_names[s]].set_allocation(alloc)
+            rec.acc_idx[s] = -2 if alloc.accelerator == "" else acc_index[alloc.accelerator]
+            rec.num_replicas[s] = alloc.num_replicas
+            rec.batch[s] = alloc.batch_size
+            rec.cost[s] = alloc.cost
+            rec.itl[s] = alloc.itl
+            rec.ttft[s] = alloc.ttft
+        return rec
 
     def _allgather_candidates(self, system: System, local_names, acc_names, world):
         """Greedy limited mode at world>1: all-gather every rank's candidate
