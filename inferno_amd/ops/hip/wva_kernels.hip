@@ -879,22 +879,32 @@ extern "C" __global__ void __launch_bounds__(WVA_WAVE) wva_argmin(
 // ---------------------------------------------------------------------------
 // runtime (nt, gmem) -> template dispatch. GMEM instantiations exist for the
 // two block widths the host's huge-bucket policy uses (256, 1024).
-// one-time >64 KiB dynamic-LDS opt-in per kernel instantiation (XL bucket);
-// opts straight into the full 160 KiB so repeat launches skip the runtime call
+// >64 KiB dynamic-LDS opt-in per kernel instantiation (XL bucket). The
+// runtime rejects opting into the full 160 KiB (measured: hipErrorInvalidValue
+// above ~the launch-bounds-derived cap), so opt into the exact size needed,
+// growing monotonically and caching per func so steady-state launches skip
+// the runtime call.
 static int wva_optin_lds(const void *func, size_t lds) {
   constexpr size_t kDefaultCap = 64 * 1024;
   if (lds <= kDefaultCap) return 0;
-  static const void *done[8] = {};
-  for (int i = 0; i < 8; ++i)
-    if (done[i] == func) return 0;
-  int rc = (int)hipFuncSetAttribute(func, hipFuncAttributeMaxDynamicSharedMemorySize,
-                                    160 * 1024);
-  if (rc != 0) return rc;
-  for (int i = 0; i < 8; ++i)
-    if (done[i] == nullptr) {
-      done[i] = func;
+  static const void *funcs[8] = {};
+  static size_t sizes[8] = {};
+  int slot = -1;
+  for (int i = 0; i < 8; ++i) {
+    if (funcs[i] == func) {
+      if (sizes[i] >= lds) return 0;
+      slot = i;
       break;
     }
+    if (funcs[i] == nullptr && slot < 0) slot = i;
+  }
+  int rc = (int)hipFuncSetAttribute(func, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                    (int)lds);
+  if (rc != 0) return rc;
+  if (slot >= 0) {
+    funcs[slot] = func;
+    sizes[slot] = lds;
+  }
   return 0;
 }
 

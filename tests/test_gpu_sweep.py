@@ -480,3 +480,22 @@ class TestHugeNGmemSpill:
         assert (diff <= 1).all() and (diff == 0).sum() >= len(diff) - 1
         np.testing.assert_array_equal(rec_cpu.batch, rec_gpu.batch)
         assert (rec_cpu.batch > 8192).any()
+
+    def test_xl_upper_bound_cell(self):
+        """N at the XL tier's ceiling (32768 -> ~136.6KB LDS): the dynamic-LDS
+        opt-in must succeed and match the CPU golden."""
+        cpu_sys, gpu_sys, opt = build_pair(n_servers=1, seed=403)
+        for s in (cpu_sys, gpu_sys):
+            for srv in s.servers.values():
+                srv.max_batch_size = 32768  # override -> N exactly at the cap
+                srv.load = ServerLoadSpec(arrivalRate=120.0, avgInTokens=64,
+                                          avgOutTokens=32)
+        SweepEngine(backend="cpu").sweep(cpu_sys)
+        SweepEngine(backend="gpu").sweep(gpu_sys)
+        for name in cpu_sys.servers:
+            a_map = cpu_sys.servers[name].all_allocations
+            b_map = gpu_sys.servers[name].all_allocations
+            assert set(a_map) == set(b_map)
+            for acc in a_map:
+                assert a_map[acc].batch_size == 32768
+                assert_alloc_close(a_map[acc], b_map[acc], name, acc)
