@@ -23,6 +23,10 @@ import statistics
 import sys
 import time
 
+# one HW queue per sweep bucket stream so the overlapped bucket launches
+# don't serialize (ROCm defaults to 4 queues; must be set before HIP init)
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 
 def _parse_args():
     p = argparse.ArgumentParser()

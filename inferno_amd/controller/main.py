@@ -15,6 +15,10 @@ import socket
 import threading
 import time
 
+# one HW queue per sweep bucket stream (ROCm defaults to 4; set before HIP
+# runtime init so the overlapped bucket launches don't serialize)
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 from ..utils.logging import init_logger
 from . import collector
 from .k8s import HttpKube

@@ -96,6 +96,13 @@ def choose_buckets(batch_n):
             nt = int(env)
         ids = None if count == n else idx.astype(np.int32)
         out.append((nt, ids, int(batch_n[mask].max()), count, False))
+    # launch order = descending max_n (most expensive bucket first): ROCm
+    # multiplexes streams onto GPU_MAX_HW_QUEUES (default 4) hardware queues,
+    # so with 5 buckets the later streams share a queue and serialize —
+    # measured: the XL bucket queued behind the large bucket, costing ~250us
+    # of lost overlap per reconcile. Put the stragglers on distinct queues
+    # and let the cheap buckets share.
+    out.sort(key=lambda b: -b[2])
     return out
 
 
