@@ -57,3 +57,14 @@ docker-build:
 
 lint:
 	$(PYTHON) -m compileall -q inferno_amd bench.py __graft_entry__.py
+
+## AddressSanitizer pass over the native C++ greedy (host code)
+asan-greedy:
+	$(PYTHON) scripts/greedy_asan_check.py
+
+## One-command local stack (apiserver stand-in + emulator + TLS prom + controller)
+stack:
+	bash deploy/install.sh --local
+
+stack-smoke:
+	bash deploy/install.sh --local --smoke

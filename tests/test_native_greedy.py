@@ -181,3 +181,37 @@ class TestNativeGreedyDifferential:
                                * acc.multiplicity)
         for t in cap:
             assert used[t] <= cap[t], t
+
+
+class TestNativeGreedyFuzz:
+    """Wider randomized differential: many fleets/capacity regimes in one
+    test (cheap: CPU sweep reused per seed)."""
+
+    @pytest.mark.parametrize("seed", range(3, 13))
+    def test_random_regimes(self, seed):
+        _require_lib()
+        rng = np.random.default_rng(1000 + seed)
+        policy = POLICIES[seed % 4]
+        cap = {
+            "AMD-MI300X-192GB": int(rng.integers(0, 50)),
+            "AMD-MI325X-256GB": int(rng.integers(0, 50)),
+            "AMD-MI355X-288GB": int(rng.integers(0, 50)),
+        }
+        kw = dict(n_servers=int(rng.integers(4, 20)), seed=2000 + seed,
+                  unlimited=False, capacity=dict(cap), saturation_policy=policy,
+                  delayed_best_effort=bool(seed % 2),
+                  priorities=(1, 5, 10) if seed % 3 == 0 else (1, 10))
+        a, opt = System.from_spec(make_spec(**kw))
+        b, _ = System.from_spec(make_spec(**kw))
+        local_names = sorted(a.servers)
+        SweepEngine(backend="cpu").sweep(b)
+        solve_greedy(b, delayed_best_effort=opt.delayedBestEffort,
+                     saturation_policy=SaturationPolicy.parse(opt.saturationPolicy))
+        fs, cells = cells_from_cpu_sweep(a, local_names)
+        run_native(a, opt, cells, fs, local_names)
+        for name in local_names:
+            ga, gb = a.servers[name].allocation, b.servers[name].allocation
+            assert (ga is None) == (gb is None), f"{name} ({policy})"
+            if ga is not None:
+                assert (ga.accelerator, ga.num_replicas) == (
+                    gb.accelerator, gb.num_replicas), f"{name} ({policy})"
