@@ -157,6 +157,7 @@ def world(tmp_path_factory):
         yield {
             "kube": kube, "emu": emu,
             "metrics_port": metrics_port, "api_port": api_port,
+            "prom_url": f"https://127.0.0.1:{prom_port}", "prom_ca": crt,
         }
         kube.close()
     finally:
@@ -317,3 +318,67 @@ class TestControllerEndToEnd:
             return r.json() if conds.get("OptimizationReady") == "True" else None
 
         _wait_for(second_optimized, 45, desc="second VA optimized")
+
+
+@pytest.mark.e2e
+class TestSecuredMetricsE2E:
+    def test_controller_serves_https_metrics_with_token_auth(self, world, tmp_path):
+        """A second controller instance with --metrics-cert-dir and a static
+        token file: /metrics requires the bearer token over HTTPS
+        (cmd/main.go:122-199 parity through the real process)."""
+        crt, key = _mk_cert(tmp_path)
+        cert_dir = tmp_path / "certs"
+        cert_dir.mkdir()
+        (cert_dir / "tls.crt").write_bytes(open(crt, "rb").read())
+        (cert_dir / "tls.key").write_bytes(open(key, "rb").read())
+        tok = tmp_path / "token"
+        tok.write_text("s3cret-metrics\n")
+
+        metrics_port = _free_port()
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "inferno_amd.controller.main",
+             "--metrics-port", str(metrics_port),
+             "--metrics-cert-dir", str(cert_dir),
+             "--metrics-auth-token-file", str(tok),
+             "--no-leader-elect", "--backend", "cpu"],
+            env={**os.environ,
+                 "KUBE_API_URL": f"http://127.0.0.1:{world['api_port']}",
+                 "PROMETHEUS_BASE_URL": world["prom_url"],
+                 "PROMETHEUS_CA_CERT_PATH": world["prom_ca"]},
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, cwd=REPO,
+        )
+        try:
+            base = f"https://127.0.0.1:{metrics_port}"
+            import ssl
+
+            ctx = ssl.create_default_context()
+            ctx.check_hostname = False
+            ctx.verify_mode = ssl.CERT_NONE
+            deadline = time.time() + 30
+            up = False
+            while time.time() < deadline:
+                if proc.poll() is not None:
+                    pytest.fail("secured controller exited early")
+                try:
+                    r = httpx.get(f"{base}/healthz", verify=ctx, timeout=2)
+                    if r.status_code == 200:
+                        up = True
+                        break
+                except httpx.HTTPError:
+                    time.sleep(0.3)
+            assert up, "secured controller probes never came up"
+            # unauthenticated scrape rejected; token accepted
+            assert httpx.get(f"{base}/metrics", verify=ctx,
+                             timeout=5).status_code == 401
+            r = httpx.get(f"{base}/metrics", verify=ctx, timeout=5,
+                          headers={"Authorization": "Bearer s3cret-metrics"})
+            assert r.status_code == 200
+            # plain HTTP against the TLS port fails
+            with pytest.raises(httpx.HTTPError):
+                httpx.get(f"http://127.0.0.1:{metrics_port}/healthz", timeout=2)
+        finally:
+            proc.send_signal(signal.SIGTERM)
+            try:
+                proc.wait(timeout=10)
+            except subprocess.TimeoutExpired:
+                proc.kill()
