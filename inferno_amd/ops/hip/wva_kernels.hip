@@ -222,17 +222,13 @@ struct ChainOut {
 // Each lane owns a CONTIGUOUS chunk of states n in [n0, n1]; within the
 // chunk, probabilities advance by the linear-space recurrence
 // w(n+1) = w(n) * lam / s(n+1) — one fp64 exp per SUB-state anchor instead
-// of one per state (the exp is the dominant per-state cost; anchors are
-// pinned <= exp(0) by the anchor max).
+// of one per state (the exp is the dominant per-state cost; a 32-state
+// sub-chunk product spans at most e^224 relative to its anchor, far inside
+// fp64 range, and anchors are pinned <= exp(0) by the anchor max).
 // LDS layout: inv_s_t is chunk-TRANSPOSED ([j*NT + lane] holds 1/s(n0+j))
 // so the per-step reads are conflict-free; S_anchor holds the log-prefix at
 // the anchor states only.
-// 16-state sub-chunks: anchors every 16 states give every lane with chunk
-// >= 32 at least TWO independent recurrence chains, which pass 2 interleaves
-// to hide the fp64 FMA+LDS latency of the serial w-chain (the measured
-// per-eval bound). The fp32 reciprocal product within a sub-chunk spans at
-// most ~e^112 relative to its anchor — far inside fp64 range.
-#define WVA_SUB 16
+#define WVA_SUB 32
 
 struct ChainGeom {
   const float *inv_s_t;    // [chunk*NT] transposed reciprocal service rates
@@ -317,67 +313,13 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
   }
   const double m = red_max_p<NT, PART>(tmax, scratch);
 
-  // pass 2: head sums via per-sub-chunk running products. Sub-chunks are
-  // INDEPENDENT (each starts from its own anchor exp), so pairs of them run
-  // interleaved with separate accumulators — two in-flight fp64 recurrences
-  // hide each other's FMA+LDS latency (the serial w-chain is the measured
-  // per-eval bound: PMC wait_frac ~0.5, cyc/wave tracking chunk length).
-  // Per-chain accumulation order is unchanged; only the cross-sub-chunk
-  // addition order differs from the sequential walk (fp64 reorder ~1e-15,
-  // far inside the 1e-6 bisection tolerance).
+  // pass 2: head sums via per-sub-chunk running products
   double head_sum = (lane == 0) ? exp(-m) : 0.0;  // n = 0 state
   double head_n_sum = 0.0;
   for (int c = lane; c < NT; c += W) {
     const int n0c = c * g.chunk + 1;
     const int n1c = min(n0c + g.chunk - 1, N);
-    int k = 0;
-    // paired sub-chunks: both anchors in range and chain 0 full-length
-    for (; k + 1 < g.ksub; k += 2) {
-      const int ja = k * WVA_SUB;
-      const int jb = ja + WVA_SUB;
-      const int na = n0c + ja;
-      const int nb = n0c + jb;
-      if (nb > n1c) break;  // chain b absent/partial: fall to singles
-      double wa = exp((double)na * loglam - g.S_anchor[c * g.ksub + k] - m);
-      double wb = exp((double)nb * loglam - g.S_anchor[c * g.ksub + k + 1] - m);
-      double sa = wa, sb = wb;
-      double nsa = (double)na * wa, nsb = (double)nb * wb;
-      const int ea = jb - 1;  // chain a is full: WVA_SUB-1 recurrence steps
-      const int eb = min(jb + WVA_SUB - 1, n1c - n0c);
-      int ji = ja + 1, jj = jb + 1;
-      for (; ji + 1 <= ea && jj + 1 <= eb; ji += 2, jj += 2) {
-        const double fa0 = (double)g.inv_s_t[ji * NT + c];
-        const double fa1 = (double)g.inv_s_t[(ji + 1) * NT + c];
-        const double fb0 = (double)g.inv_s_t[jj * NT + c];
-        const double fb1 = (double)g.inv_s_t[(jj + 1) * NT + c];
-        wa *= lam * fa0;
-        sa += wa;
-        nsa += (double)(n0c + ji) * wa;
-        wb *= lam * fb0;
-        sb += wb;
-        nsb += (double)(n0c + jj) * wb;
-        wa *= lam * fa1;
-        sa += wa;
-        nsa += (double)(n0c + ji + 1) * wa;
-        wb *= lam * fb1;
-        sb += wb;
-        nsb += (double)(n0c + jj + 1) * wb;
-      }
-      for (; ji <= ea; ++ji) {
-        wa *= lam * (double)g.inv_s_t[ji * NT + c];
-        sa += wa;
-        nsa += (double)(n0c + ji) * wa;
-      }
-      for (; jj <= eb; ++jj) {
-        wb *= lam * (double)g.inv_s_t[jj * NT + c];
-        sb += wb;
-        nsb += (double)(n0c + jj) * wb;
-      }
-      head_sum += sa + sb;
-      head_n_sum += nsa + nsb;
-    }
-    // remaining single (or partial) sub-chunks — 4-wide batched LDS reads
-    for (; k < g.ksub; ++k) {
+    for (int k = 0; k < g.ksub; ++k) {
       const int j0 = k * WVA_SUB;
       const int na = n0c + j0;
       if (na > n1c) break;
@@ -385,6 +327,11 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
       head_sum += w;
       head_n_sum += (double)na * w;
       const int jend = min(j0 + WVA_SUB - 1, n1c - n0c);
+      // batch the LDS reads 4-wide ahead of the dependent w-chain: with the
+      // narrow-block dispatch there is often only one wave per SIMD, so an
+      // un-batched loop exposes the full LDS latency on every step (PMC:
+      // SQ_WAIT_ANY was 50% of wave cycles). Arithmetic order is unchanged
+      // (bit-identical results).
       int j = j0 + 1;
       for (; j + 3 <= jend; j += 4) {
         const double f0 = (double)g.inv_s_t[j * NT + c];
@@ -971,7 +918,7 @@ static int wva_sweep_dispatch(int n_blocks, int max_n, int nt, const int *cell_i
   if (!gmem && max_n > WVA_XL_MAX_N) return -2;
   if (gmem && max_n > WVA_HUGE_MAX_N) return -4;
   const int chunk = (max_n + nt - 1) / nt;
-  const int ksub = (chunk + WVA_SUB - 1) / WVA_SUB;
+  const int ksub = (chunk + 32 - 1) / 32;  // WVA_SUB
   // header(40) + inv_s floats (chunk*nt/2 doubles, rounded up) + anchors;
   // the spill path keeps only the header in LDS
   size_t lds = gmem ? (size_t)40 * sizeof(double)

@@ -113,7 +113,7 @@ def alloc_gmem_slabs(n_blocks: int, max_n: int, nt: int, device: str):
     import torch
 
     chunk = (max_n + nt - 1) // nt
-    ksub = (chunk + 15) // 16  # WVA_SUB=16 anchors
+    ksub = (chunk + 31) // 32
     g_inv = torch.empty((n_blocks, chunk * nt), dtype=torch.float32, device=device)
     g_anchor = torch.empty((n_blocks, nt * ksub), dtype=torch.float64, device=device)
     return g_inv, g_anchor
