@@ -45,6 +45,7 @@
 #include <hip/hip_runtime.h>
 #include <math.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #define WVA_WAVE 64
 // max LDS-resident batch size per cell (bounds the LDS chain geometry:
@@ -1072,6 +1073,12 @@ struct WvaCtx {
   hipStream_t side[WVA_MAX_BUCKETS - 1];    // overlap streams for buckets 1..
   hipEvent_t e_up;                          // upload-complete
   hipEvent_t e_b[WVA_MAX_BUCKETS - 1];      // side-bucket completion
+  // hipGraph capture of the whole per-tick pipeline (H2D copies, bucket
+  // kernels on their streams, argmin+gather, D2H): replayed as ONE launch
+  // per reconcile. State: 0 = not captured, 1 = exec valid, -1 = capture
+  // failed once -> stay on the eager path. Invalidated by set_buckets.
+  hipGraphExec_t graph_exec;
+  int graph_state;
 };
 
 // pointer-slot order for wva_ctx_create (host mirrors in ops/sweep.py):
@@ -1091,6 +1098,8 @@ extern "C" void *wva_ctx_create(int n_cells, int n_srv, void **p) {
   c->analyzer_mode = 0;
   c->cv2 = 1.0f;
   c->n_buckets = 0;
+  c->graph_exec = nullptr;
+  c->graph_state = 0;
   char *di = (char *)p[0];
   char *df = (char *)p[1];
   size_t ci = (size_t)n_cells * sizeof(int);
@@ -1162,6 +1171,10 @@ extern "C" int wva_ctx_set_buckets(void *ctx, int n_buckets, const int *nts,
   }
   c->analyzer_mode = analyzer_mode;
   c->cv2 = cv2;
+  // bucket layout / analyzer change: drop the captured pipeline (unless
+  // capture already proved unsupported, which is sticky)
+  if (c->graph_state == 1) (void)hipGraphExecDestroy(c->graph_exec);
+  if (c->graph_state != -1) c->graph_state = 0;
   return 0;
 }
 
@@ -1170,8 +1183,9 @@ static int wva_launch_bucket_on(WvaCtx *c, const WvaBucket &b, hipStream_t s) {
                             c->in, c->out, b.g_inv, b.g_anchor);
 }
 
-extern "C" int wva_reconcile(void *ctx) {
-  WvaCtx *c = (WvaCtx *)ctx;
+// enqueue the whole per-tick pipeline on the ctx's streams (used both for
+// eager execution and for stream capture into a hipGraph)
+static int wva_enqueue_pipeline(WvaCtx *c) {
   hipError_t err;
   // 1. dynamic H2D (pinned -> device)
   err = hipMemcpyAsync(c->dev_i, c->pin_i, (size_t)6 * c->n_cells * sizeof(int),
@@ -1204,19 +1218,75 @@ extern "C" int wva_reconcile(void *ctx) {
   err = hipGetLastError();
   if (err != hipSuccess) return (int)err;
 
-  // 4. D2H of the winner records + full sync
+  // 4. D2H of the winner records
   err = hipMemcpyAsync(c->pin_out_f, c->gather_f, (size_t)6 * c->n_srv * sizeof(float),
                        hipMemcpyDeviceToHost, c->s0);
   if (err != hipSuccess) return (int)err;
   err = hipMemcpyAsync(c->pin_out_i, c->gather_i, (size_t)4 * c->n_srv * sizeof(int),
                        hipMemcpyDeviceToHost, c->s0);
   if (err != hipSuccess) return (int)err;
+  return 0;
+}
+
+extern "C" int wva_reconcile(void *ctx) {
+  WvaCtx *c = (WvaCtx *)ctx;
+  // replay the captured pipeline when available: every buffer pointer and
+  // kernel argument is ctx-stable between ticks, so the graph stays valid
+  // until the bucket layout changes (set_buckets invalidates)
+  if (c->graph_state == 1) {
+    if (hipGraphLaunch(c->graph_exec, c->s0) != hipSuccess) {
+      (void)hipGraphExecDestroy(c->graph_exec);
+      c->graph_state = -1;  // fall back to eager permanently
+    } else {
+      return (int)hipStreamSynchronize(c->s0);
+    }
+  }
+  if (c->graph_state == 0) {
+    static int disabled = -1;
+    if (disabled < 0) disabled = getenv("INFERNO_NO_HIPGRAPH") != nullptr ? 1 : 0;
+    if (disabled) c->graph_state = -1;
+  }
+  if (c->graph_state == 0) {
+    // capture once: the side-stream forks/joins via events become graph
+    // dependencies (capture mode relaxed — no other thread uses these
+    // streams)
+    hipGraph_t graph = nullptr;
+    bool ok = hipStreamBeginCapture(c->s0, hipStreamCaptureModeRelaxed) == hipSuccess;
+    int rc = ok ? wva_enqueue_pipeline(c) : -20;
+    if (ok) {
+      if (hipStreamEndCapture(c->s0, &graph) != hipSuccess) ok = false;
+    }
+    if (ok && rc == 0 && graph != nullptr &&
+        hipGraphInstantiate(&c->graph_exec, graph, nullptr, nullptr, 0) == hipSuccess) {
+      c->graph_state = 1;
+      (void)hipGraphDestroy(graph);
+      if (hipGraphLaunch(c->graph_exec, c->s0) == hipSuccess)
+        return (int)hipStreamSynchronize(c->s0);
+      (void)hipGraphExecDestroy(c->graph_exec);
+      c->graph_state = -1;
+    } else {
+      if (graph != nullptr) (void)hipGraphDestroy(graph);
+      c->graph_state = -1;  // capture unsupported here: stay eager
+      // a failed capture may have left the stream in capture state; make
+      // sure it is usable again
+      hipStreamCaptureStatus st;
+      if (hipStreamIsCapturing(c->s0, &st) == hipSuccess &&
+          st != hipStreamCaptureStatusNone) {
+        hipGraph_t dead = nullptr;
+        (void)hipStreamEndCapture(c->s0, &dead);
+        if (dead != nullptr) (void)hipGraphDestroy(dead);
+      }
+    }
+  }
+  int rc = wva_enqueue_pipeline(c);
+  if (rc != 0) return rc;
   return (int)hipStreamSynchronize(c->s0);
 }
 
 extern "C" void wva_ctx_destroy(void *ctx) {
   WvaCtx *c = (WvaCtx *)ctx;
   if (c == nullptr) return;
+  if (c->graph_state == 1) (void)hipGraphExecDestroy(c->graph_exec);
   (void)hipStreamDestroy(c->s0);
   (void)hipEventDestroy(c->e_up);
   for (int i = 0; i < WVA_MAX_BUCKETS - 1; ++i) {
