@@ -1242,9 +1242,13 @@ extern "C" int wva_reconcile(void *ctx) {
     }
   }
   if (c->graph_state == 0) {
-    static int disabled = -1;
-    if (disabled < 0) disabled = getenv("INFERNO_NO_HIPGRAPH") != nullptr ? 1 : 0;
-    if (disabled) c->graph_state = -1;
+    // OPT-IN (INFERNO_HIPGRAPH=1): measured SLOWER than eager on MI355X for
+    // this pipeline (0.55 vs 0.45 ms/step, same-box x3) — graph replay does
+    // not preserve the 4-HW-queue bucket overlap the eager path gets. Kept
+    // as an option for launch-bound configurations (many tiny buckets).
+    static int enabled = -1;
+    if (enabled < 0) enabled = getenv("INFERNO_HIPGRAPH") != nullptr ? 1 : 0;
+    if (!enabled) c->graph_state = -1;
   }
   if (c->graph_state == 0) {
     // capture once: the side-stream forks/joins via events become graph
