@@ -21,6 +21,9 @@ def main() -> None:
     p.add_argument("--fleets", type=int, default=40)
     p.add_argument("--servers", type=int, default=32)
     p.add_argument("--seed0", type=int, default=10_000)
+    p.add_argument("--huge", action="store_true",
+                   help="push every 4th server into the XL/GMEM N tiers "
+                        "(uncapped batch states, round-2 spill paths)")
     args = p.parse_args()
 
     from inferno_amd.core import System
@@ -50,6 +53,24 @@ def main() -> None:
         sb.optimizer.analyzer = analyzer
         a, opt = System.from_spec(sa)
         b, _ = System.from_spec(sb)
+        if args.huge:
+            # force a slice of servers into the XL (8192<N<=32768) and GMEM
+            # (N>32768) geometry tiers: tiny K with big maxBatchSize*atTokens
+            import numpy as _np
+
+            hrng = _np.random.default_rng(seed ^ 0xBEEF)
+            for s in (a, b):
+                srng = _np.random.default_rng(seed ^ 0xBEEF)  # same per system
+                for i, srv_name in enumerate(sorted(s.servers)):
+                    if i % 4 != 0:
+                        continue
+                    srv = s.servers[srv_name]
+                    k_out = int(srng.integers(5, 40))
+                    if srv.load is not None and srv.load.arrivalRate > 0:
+                        srv.load.avgOutTokens = k_out
+                    for perf in s.models[srv.model_name].perf_data.values():
+                        perf.maxBatchSize = 256
+                        perf.atTokens = 2048
         SweepEngine(backend="cpu").sweep(a)
         SweepEngine(backend="gpu").sweep(b)
         for name in a.servers:
