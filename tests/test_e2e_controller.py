@@ -382,3 +382,54 @@ class TestSecuredMetricsE2E:
                 proc.wait(timeout=10)
             except subprocess.TimeoutExpired:
                 proc.kill()
+
+
+@pytest.mark.e2e
+class TestTpVariantE2E:
+    """The MI355X TP-variant path end to end: the llama70b example VA
+    (perfmodel-derived MI355X/TP4/TP8 profiles, deploy/examples) reconciles
+    through the real controller — TP degree as a first-class variant axis
+    priced via the xGMI all-reduce model (BASELINE config 3 shape)."""
+
+    def test_llama70b_tp_variants_optimized(self, world):
+        kube = world["kube"]
+        assert _apply(kube, {
+            "apiVersion": "apps/v1", "kind": "Deployment",
+            "metadata": {"name": "llama70b-deploy", "namespace": VA_NS},
+            "spec": {"replicas": 1}, "status": {"replicas": 1},
+        }).status_code == 201
+        with open(os.path.join(REPO, "deploy", "examples",
+                               "vllme-variantautoscaling.yaml")) as f:
+            va_doc = next(d for d in yaml.safe_load_all(f)
+                          if d and d["metadata"]["name"] == "llama70b-deploy")
+        va_doc["metadata"]["namespace"] = VA_NS
+        # the emulator exports metrics for default/default only; point the
+        # llama VA's modelID at it so the collector finds live load (the TP
+        # profiles under test are unchanged)
+        va_doc["spec"]["modelID"] = MODEL
+        # keepAccelerator pins candidates to the labeled accelerator; label
+        # the TP4 variant so the sweep prices accCount=4 over xGMI
+        va_doc["metadata"]["labels"][
+            "inference.optimization/acceleratorName"] = "MI355X-TP4"
+        r = _apply(kube, va_doc)
+        assert r.status_code == 201, r.text
+
+        _drive_load(world["emu"], seconds=6.0, concurrency=4)
+
+        def optimized():
+            resp = kube.get(
+                f"/apis/llmd.ai/v1alpha1/namespaces/{VA_NS}"
+                "/variantautoscalings/llama70b-deploy")
+            if resp.status_code != 200:
+                return None
+            st = resp.json().get("status", {})
+            conds = {c["type"]: c["status"] for c in st.get("conditions", [])}
+            des = st.get("desiredOptimizedAlloc", {})
+            if conds.get("OptimizationReady") == "True" and des.get(
+                    "numReplicas", 0) >= 1:
+                return st
+            return None
+
+        st = _wait_for(optimized, 45, desc="llama70b TP-variant optimization")
+        # keepAccelerator restricts the winner to the labeled TP variant
+        assert st["desiredOptimizedAlloc"]["accelerator"] == "MI355X-TP4"
