@@ -323,3 +323,66 @@ class TestLeaderElectionE2E:
         # A stops renewing; after expiry B takes over
         time.sleep(2.5)
         assert b.try_acquire() is True
+
+
+class TestMergePatchProperties:
+    """RFC 7386 merge-patch semantics of the apiserver stand-in, checked
+    against an independent reference implementation over randomized docs."""
+
+    @staticmethod
+    def _rfc7386(target, patch):
+        # independent re-implementation straight from the RFC pseudocode
+        if not isinstance(patch, dict):
+            return patch
+        if not isinstance(target, dict):
+            target = {}
+        result = dict(target)
+        for k, v in patch.items():
+            if v is None:
+                result.pop(k, None)
+            else:
+                result[k] = TestMergePatchProperties._rfc7386(result.get(k), v)
+        return result
+
+    def test_rfc_examples(self):
+        from inferno_amd.testing.kubeapi import merge_patch
+
+        cases = [
+            ({"a": "b"}, {"a": "c"}, {"a": "c"}),
+            ({"a": "b"}, {"b": "c"}, {"a": "b", "b": "c"}),
+            ({"a": "b"}, {"a": None}, {}),
+            ({"a": "b", "b": "c"}, {"a": None}, {"b": "c"}),
+            ({"a": ["b"]}, {"a": "c"}, {"a": "c"}),
+            ({"a": "c"}, {"a": ["b"]}, {"a": ["b"]}),
+            ({"a": {"b": "c"}}, {"a": {"b": "d", "c": None}}, {"a": {"b": "d"}}),
+            ({"a": [{"b": "c"}]}, {"a": [1]}, {"a": [1]}),
+            (["a", "b"], ["c", "d"], ["c", "d"]),
+            ({"a": "b"}, ["c"], ["c"]),
+            ({"a": "foo"}, None, None),
+            ({"a": "foo"}, "bar", "bar"),
+            ({"e": None}, {"a": 1}, {"e": None, "a": 1}),
+            ([1, 2], {"a": "b", "c": None}, {"a": "b"}),
+            ({}, {"a": {"bb": {"ccc": None}}}, {"a": {"bb": {}}}),
+        ]
+        for target, patch, want in cases:
+            assert merge_patch(target, patch) == want, (target, patch)
+
+    def test_randomized_against_reference(self):
+        import random
+
+        from inferno_amd.testing.kubeapi import merge_patch
+
+        rng = random.Random(99)
+
+        def rand_doc(depth=0):
+            r = rng.random()
+            if depth > 3 or r < 0.25:
+                return rng.choice([None, 1, "x", True, [1, 2], "y"])
+            return {
+                rng.choice("abcde"): rand_doc(depth + 1)
+                for _ in range(rng.randint(0, 4))
+            }
+
+        for _ in range(300):
+            t, p = rand_doc(), rand_doc()
+            assert merge_patch(t, p) == self._rfc7386(t, p)
