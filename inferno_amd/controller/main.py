@@ -153,7 +153,14 @@ def main() -> None:
             time.sleep(2.0)
             continue
         wake.clear()
-        result = reconciler.reconcile()
+        try:
+            result = reconciler.reconcile()
+        except Exception as e:  # noqa: BLE001 - last-resort guard: a tick
+            # must never kill the process (transient API/Prometheus/network
+            # failures degrade to an errored tick + requeue)
+            logger.error("reconcile tick failed", extra={"kv": {"err": str(e)}})
+            stop.wait(5.0)
+            continue
         logger.info(
             "reconcile complete",
             extra={

@@ -153,7 +153,12 @@ class Reconciler:
 
     # ------------------------------------------------------------------
     def read_interval(self) -> float:
-        cm = self.kube.get_configmap(self.configmap_namespace, WVA_CONFIG_CM) or {}
+        try:
+            cm = self.kube.get_configmap(self.configmap_namespace, WVA_CONFIG_CM) or {}
+        except Exception:
+            # API unreachable: keep the default requeue cadence (the tick
+            # itself will record the error; ref requeue-on-error semantics)
+            return DEFAULT_INTERVAL_SECONDS
         interval = cm.get("GLOBAL_OPT_INTERVAL", "")
         if interval:
             try:
@@ -209,16 +214,24 @@ class Reconciler:
         self._maybe_reprobe_gpu()
         result = ReconcileResult(requeue_after=self.read_interval())
 
-        accelerator_cm = self._read_accelerator_cm()
-        if accelerator_cm is None:
-            result.errors.append("unable to read accelerator configMap")
-            return self._finish(result, t_start)
-        service_class_cm = self._read_service_class_cm()
-        if service_class_cm is None:
-            result.errors.append("unable to read serviceclass configMap")
-            return self._finish(result, t_start)
+        # API-server errors on the tick's list/ConfigMap reads must degrade
+        # to a failed tick, never crash the process (ref: controller-runtime
+        # logs and requeues on client errors)
+        try:
+            accelerator_cm = self._read_accelerator_cm()
+            if accelerator_cm is None:
+                result.errors.append("unable to read accelerator configMap")
+                return self._finish(result, t_start)
+            service_class_cm = self._read_service_class_cm()
+            if service_class_cm is None:
+                result.errors.append("unable to read serviceclass configMap")
+                return self._finish(result, t_start)
 
-        vas = [va for va in self.kube.list_variantautoscalings() if not va.deletionTimestamp]
+            vas = [va for va in self.kube.list_variantautoscalings()
+                   if not va.deletionTimestamp]
+        except Exception as e:
+            result.errors.append(f"API server unreachable: {e}")
+            return self._finish(result, t_start)
         if not vas:
             return self._finish(result, t_start)
 
@@ -278,10 +291,16 @@ class Reconciler:
             va.status.currentAlloc = current_alloc
             return (va, class_name)
 
+        def _collect_safe(va: api.VariantAutoscaling):
+            try:
+                return _collect(va)
+            except Exception as e:  # noqa: BLE001 - per-VA continue-on-error
+                return f"{va.name}: prepare failed: {e}"
+
         if self.collect_workers > 1 and len(vas) > 1:
-            collected = list(self._pool().map(_collect, vas))
+            collected = list(self._pool().map(_collect_safe, vas))
         else:
-            collected = [_collect(va) for va in vas]
+            collected = [_collect_safe(va) for va in vas]
 
         update_list: list[api.VariantAutoscaling] = []
         for va, item in zip(vas, collected):
