@@ -249,3 +249,28 @@ def create_optimized_alloc(
 def get_config_value(data: dict[str, str], key: str, default: str) -> str:
     """Ref utils.go GetConfigValue."""
     return data.get(key, default)
+
+
+def capacity_from_inventory(
+    spec: SystemSpec, inventory: dict[str, dict[str, dict]]
+) -> list:
+    """Map the cluster GPU inventory (collector.collect_inventory_k8s shape:
+    {vendor: {product: {count, memory}}}) onto the SystemSpec's accelerator
+    TYPES for the limited-mode solver's capacity constraint.
+
+    A node's ``gpu.product`` matches an accelerator when it equals the
+    accelerator's name (the unit-cost ConfigMap key, e.g. "MI355X") or its
+    type/device string (e.g. "AMD-MI355X-288GB"). Unmatched products are
+    ignored (no accelerator profile to price them)."""
+    from ..config import AcceleratorCount
+
+    per_type: dict[str, int] = {}
+    for _vendor, products in (inventory or {}).items():
+        for product, info in products.items():
+            for acc in spec.accelerators:
+                if product == acc.name or product == acc.type:
+                    per_type[acc.type] = per_type.get(acc.type, 0) + int(
+                        info.get("count", 0)
+                    )
+                    break
+    return [AcceleratorCount(type=t, count=c) for t, c in sorted(per_type.items())]

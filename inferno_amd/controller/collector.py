@@ -287,11 +287,52 @@ def add_metrics_to_opt_status(
     )
 
 
+# GPU vendor label prefixes (ref collector.go:31-35)
+GPU_VENDORS = ("nvidia.com", "amd.com", "intel.com")
+
+
 def collect_inventory_k8s(kube=None) -> dict[str, dict[str, dict]]:
-    """Stub for future limited-mode support — WVA operates in unlimited mode
-    (ref internal/collector/collector.go:37-42 CollectInventoryK8S). The GPU
-    vendor label prefixes it will scan: nvidia.com, amd.com, intel.com."""
-    return {}
+    """Cluster GPU inventory from node labels — the real implementation of
+    the reference's declared-TODO stub (collector.go:37-42 CollectInventoryK8S,
+    "will be properly implemented for limited mode").
+
+    Scans nodes for the reference's emulated-GPU label convention
+    (deploy/kind-emulator/setup.sh:120-133):
+        {vendor}.com/gpu.count   "4"
+        {vendor}.com/gpu.product "MI355X"
+        {vendor}.com/gpu.memory  "288GB"
+    Returns {vendor: {product: {"count": int, "memory": str}}} — the shape of
+    the reference's map[string]map[string]AcceleratorModelInfo. Capacity for
+    the limited-mode solver derives from this via
+    ``adapters.capacity_from_inventory``."""
+    if kube is None or not hasattr(kube, "list_nodes"):
+        return {}
+    inventory: dict[str, dict[str, dict]] = {}
+    try:
+        nodes = kube.list_nodes()
+    except Exception:
+        return {}
+    for node in nodes:
+        labels = node.labels or {}
+        for vendor in GPU_VENDORS:
+            count_s = labels.get(f"{vendor}/gpu.count", "")
+            if not count_s:
+                continue
+            try:
+                count = int(count_s)
+            except ValueError:
+                continue
+            product = labels.get(f"{vendor}/gpu.product", "")
+            memory = labels.get(f"{vendor}/gpu.memory", "")
+            if count <= 0 or not product:
+                continue
+            entry = inventory.setdefault(vendor, {}).setdefault(
+                product, {"count": 0, "memory": memory}
+            )
+            entry["count"] += count
+            if memory and not entry["memory"]:
+                entry["memory"] = memory
+    return inventory
 
 
 class MockPromAPI:

@@ -45,6 +45,17 @@ class KubeClient(Protocol):  # pragma: no cover - protocol
     def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None: ...
 
 
+@dataclass
+class Node:
+    """Cluster node as the inventory collector sees it: name + labels
+    (the reference's emulated-GPU convention labels nodes with
+    ``{vendor}.com/gpu.count|.product|.memory``, deploy/kind-emulator/
+    setup.sh:120-133)."""
+
+    name: str
+    labels: dict[str, str] = field(default_factory=dict)
+
+
 class InMemoryKube:
     """In-memory fake of the API server (test double)."""
 
@@ -52,6 +63,7 @@ class InMemoryKube:
         self.vas: dict[tuple[str, str], api.VariantAutoscaling] = {}
         self.configmaps: dict[tuple[str, str], dict[str, str]] = {}
         self.deployments: dict[tuple[str, str], Deployment] = {}
+        self.nodes: dict[str, Node] = {}
         self.status_updates: list[dict[str, Any]] = []
         self._rv_counter = 0
 
@@ -70,6 +82,12 @@ class InMemoryKube:
 
     def add_deployment(self, deploy: Deployment) -> None:
         self.deployments[(deploy.namespace, deploy.name)] = deploy
+
+    def add_node(self, node: Node) -> None:
+        self.nodes[node.name] = node
+
+    def list_nodes(self) -> list[Node]:
+        return [copy.deepcopy(n) for n in self.nodes.values()]
 
     # -- KubeClient ----------------------------------------------------
     def list_variantautoscalings(self) -> list[api.VariantAutoscaling]:
@@ -190,6 +208,17 @@ class HttpKube:
     def get_configmap(self, namespace: str, name: str) -> Optional[dict[str, str]]:
         doc = self._get(f"/api/v1/namespaces/{namespace}/configmaps/{name}")
         return None if doc is None else dict(doc.get("data", {}) or {})
+
+    def list_nodes(self) -> list[Node]:
+        doc = self._get("/api/v1/nodes")
+        if doc is None:
+            return []
+        out = []
+        for item in doc.get("items", []) or []:
+            meta = item.get("metadata", {}) or {}
+            out.append(Node(name=meta.get("name", ""),
+                            labels=dict(meta.get("labels", {}) or {})))
+        return out
 
     def get_deployment(self, namespace: str, name: str) -> Optional[Deployment]:
         doc = self._get(f"/apis/apps/v1/namespaces/{namespace}/deployments/{name}")

@@ -236,6 +236,20 @@ class Reconciler:
             return self._finish(result, t_start)
 
         spec = adapters.create_system_data(accelerator_cm, service_class_cm)
+        # Opt-in limited mode (beyond the reference, which forces Unlimited
+        # and stubs inventory collection — collector.go:37-42): capacity from
+        # the cluster's GPU node labels constrains the greedy solver.
+        import os as _os
+
+        if _os.environ.get("WVA_LIMITED_MODE") == "true":
+            inventory = collector.collect_inventory_k8s(self.kube)
+            capacity = adapters.capacity_from_inventory(spec, inventory)
+            if capacity:
+                spec.optimizer.unlimited = False
+                spec.optimizer.saturationPolicy = _os.environ.get(
+                    "WVA_SATURATION_POLICY", "None"
+                )
+                spec.capacity = capacity
         import json
 
         acc_costs: dict[str, float] = {}

@@ -130,6 +130,7 @@ class ResourceKind:
 
 BUILTIN_KINDS = [
     ResourceKind("", "v1", "configmaps", "ConfigMap"),
+    ResourceKind("", "v1", "nodes", "Node", namespaced=False, has_status=True),
     ResourceKind("apps", "v1", "deployments", "Deployment", has_status=True),
     ResourceKind("coordination.k8s.io", "v1", "leases", "Lease"),
 ]
@@ -413,6 +414,15 @@ def build_app(store: Store, valid_tokens: Optional[set[str]] = None):
             if "exists" in str(e):
                 return status_err(409, "AlreadyExists", f"{plural}/{name} already exists")
             return status_err(404, "NotFound", f"{plural}/{name} not found")
+
+    # core/v1 cluster-scoped (nodes)
+    @app.api_route("/api/v1/nodes", methods=["GET", "POST"])
+    async def core_nodes(request: Request):
+        return await handle(request, "nodes", "", None)
+
+    @app.api_route("/api/v1/nodes/{name}", methods=["GET", "PUT", "PATCH", "DELETE"])
+    async def core_node_named(request: Request, name: str):
+        return await handle(request, "nodes", "", name)
 
     # core/v1
     @app.api_route("/api/v1/namespaces/{ns}/{plural}", methods=["GET", "POST"])

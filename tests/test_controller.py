@@ -620,3 +620,68 @@ class TestGpuReprobe:
         rec._maybe_reprobe_gpu()
         assert rec.engine.backend == "gpu"
         assert rec.engine.degraded_from_gpu is False
+
+
+class TestInventoryCollection:
+    """Real CollectInventoryK8S (the reference's declared-TODO stub,
+    collector.go:37-42) + opt-in limited mode fed by cluster GPU inventory."""
+
+    def test_inventory_from_node_labels(self):
+        from inferno_amd.controller.k8s import InMemoryKube, Node
+
+        kube = InMemoryKube()
+        kube.add_node(Node("n0", {"amd.com/gpu.count": "4",
+                                  "amd.com/gpu.product": "MI355X",
+                                  "amd.com/gpu.memory": "288GB"}))
+        kube.add_node(Node("n1", {"amd.com/gpu.count": "4",
+                                  "amd.com/gpu.product": "MI355X",
+                                  "amd.com/gpu.memory": "288GB"}))
+        kube.add_node(Node("n2", {"nvidia.com/gpu.count": "2",
+                                  "nvidia.com/gpu.product": "A100",
+                                  "nvidia.com/gpu.memory": "80GB"}))
+        kube.add_node(Node("n3", {"amd.com/gpu.count": "bogus",
+                                  "amd.com/gpu.product": "MI300X"}))
+        kube.add_node(Node("n4", {}))  # no GPUs
+        inv = collector.collect_inventory_k8s(kube)
+        assert inv["amd.com"]["MI355X"] == {"count": 8, "memory": "288GB"}
+        assert inv["nvidia.com"]["A100"]["count"] == 2
+        assert "MI300X" not in inv.get("amd.com", {})
+
+    def test_capacity_mapping_by_name_and_type(self):
+        spec = adapters.create_system_data(ACCELERATOR_CM, SERVICE_CLASS_CM)
+        inv = {"amd.com": {"MI355X": {"count": 8, "memory": "288GB"},
+                           "AMD-MI300X-192GB": {"count": 4, "memory": "192GB"},
+                           "UnknownChip": {"count": 9, "memory": "1GB"}}}
+        caps = adapters.capacity_from_inventory(spec, inv)
+        by_type = {c.type: c.count for c in caps}
+        assert by_type["AMD-MI355X-288GB"] == 8  # matched by NAME
+        assert by_type["AMD-MI300X-192GB"] == 4  # matched by TYPE string
+        assert "UnknownChip" not in by_type
+
+    def test_limited_mode_constrains_replicas(self, monkeypatch):
+        from inferno_amd.controller.k8s import Node
+
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=50.0)
+        # unlimited baseline needs >2 replicas under this load
+        rec.reconcile()
+        va = kube.vas[("default", "vllme-deploy")]
+        unconstrained = va.status.desiredOptimizedAlloc.numReplicas
+        assert unconstrained >= 2
+
+        kube2, prom2, em2, reg2, rec2 = build_world(arrival_per_sec=50.0)
+        kube2.add_node(Node("gpu-node", {"amd.com/gpu.count": "1",
+                                         "amd.com/gpu.product": "MI355X",
+                                         "amd.com/gpu.memory": "288GB"}))
+        monkeypatch.setenv("WVA_LIMITED_MODE", "true")
+        monkeypatch.setenv("WVA_SATURATION_POLICY", "PriorityExhaustive")
+        rec2.reconcile()
+        va2 = kube2.vas[("default", "vllme-deploy")]
+        constrained = va2.status.desiredOptimizedAlloc.numReplicas
+        assert 1 <= constrained <= 1  # capacity: 1 unit of MI355X
+        assert constrained < unconstrained
+
+    def test_limited_mode_no_nodes_falls_back_unlimited(self, monkeypatch):
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=6.0)
+        monkeypatch.setenv("WVA_LIMITED_MODE", "true")
+        result = rec.reconcile()
+        assert result.processed == 1  # no inventory -> unlimited as before
