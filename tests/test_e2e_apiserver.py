@@ -386,3 +386,31 @@ class TestMergePatchProperties:
         for _ in range(300):
             t, p = rand_doc(), rand_doc()
             assert merge_patch(t, p) == self._rfc7386(t, p)
+
+
+class TestNodeInventoryOverHttp:
+    """Cluster GPU inventory through the real HTTP apiserver stand-in
+    (nodes are cluster-scoped — a path the namespaced fakes never hit)."""
+
+    def test_list_nodes_and_collect(self, kubeapi):
+        client, port = kubeapi
+        for i, labels in enumerate([
+            {"amd.com/gpu.count": "4", "amd.com/gpu.product": "MI355X",
+             "amd.com/gpu.memory": "288GB"},
+            {"nvidia.com/gpu.count": "2", "nvidia.com/gpu.product": "A100",
+             "nvidia.com/gpu.memory": "80GB"},
+        ]):
+            r = client.post("/api/v1/nodes", json={
+                "apiVersion": "v1", "kind": "Node",
+                "metadata": {"name": f"gpu-node-{i}", "labels": labels}})
+            assert r.status_code == 201, r.text
+
+        from inferno_amd.controller import collector
+        from inferno_amd.controller.k8s import HttpKube
+
+        kube = HttpKube(base_url=f"http://127.0.0.1:{port}")
+        nodes = kube.list_nodes()
+        assert {n.name for n in nodes} >= {"gpu-node-0", "gpu-node-1"}
+        inv = collector.collect_inventory_k8s(kube)
+        assert inv["amd.com"]["MI355X"]["count"] == 4
+        assert inv["nvidia.com"]["A100"]["count"] == 2
