@@ -499,3 +499,29 @@ class TestHugeNGmemSpill:
             for acc in a_map:
                 assert a_map[acc].batch_size == 32768
                 assert_alloc_close(a_map[acc], b_map[acc], name, acc)
+
+
+class TestLimitedModeInventoryGpu:
+    def test_reconcile_limited_by_inventory_on_gpu(self, monkeypatch):
+        """WVA_LIMITED_MODE through the GPU backend: sweep on the HIP kernels,
+        capacity from node labels, winner via the native C++ greedy."""
+        from prometheus_client import CollectorRegistry
+
+        from inferno_amd.controller.k8s import Node
+        from inferno_amd.controller.metrics import MetricsEmitter
+        from inferno_amd.controller.reconciler import Reconciler
+        from tests.test_controller import build_world
+
+        kube, prom, em, reg, _ = build_world(arrival_per_sec=50.0)
+        kube.add_node(Node("gpu-node", {"amd.com/gpu.count": "1",
+                                        "amd.com/gpu.product": "MI355X",
+                                        "amd.com/gpu.memory": "288GB"}))
+        monkeypatch.setenv("WVA_LIMITED_MODE", "true")
+        monkeypatch.setenv("WVA_SATURATION_POLICY", "PriorityExhaustive")
+        rec = Reconciler(kube, prom, MetricsEmitter(registry=CollectorRegistry()),
+                         backend="gpu", scale_to_zero=False)
+        result = rec.reconcile()
+        assert result.processed == 1
+        assert result.solver_backend == "gpu"
+        va = kube.vas[("default", "vllme-deploy")]
+        assert va.status.desiredOptimizedAlloc.numReplicas == 1  # capped
