@@ -433,3 +433,46 @@ class TestTpVariantE2E:
         st = _wait_for(optimized, 45, desc="llama70b TP-variant optimization")
         # keepAccelerator restricts the winner to the labeled TP variant
         assert st["desiredOptimizedAlloc"]["accelerator"] == "MI355X-TP4"
+
+
+@pytest.mark.e2e
+class TestHpaActuationLoop:
+    """Close the actuation loop the reference's OpenShift tier asserts
+    (sharegpt_scaleup_test.go:123-214): VA recommendation -> external-metric
+    HPA (stand-in reading inferno_desired_replicas) -> Deployment replica
+    convergence -> actuator's current_replicas gauge follows."""
+
+    def test_recommendation_drives_deployment_replicas(self, world):
+        from inferno_amd.testing.hpa import read_desired, reconcile_once
+
+        kube = world["kube"]
+        _drive_load(world["emu"], seconds=8.0, concurrency=8)
+
+        metrics_url = f"http://127.0.0.1:{world['metrics_port']}/metrics"
+
+        def desired_above_one():
+            text = httpx.get(metrics_url, timeout=5).text
+            d = read_desired(text).get((VA_NS, VA_NAME), 0)
+            return d if d >= 2 else None
+
+        desired = _wait_for(desired_above_one, 60, desc="scale-out recommendation")
+
+        # one HPA pass applies the recommendation to the Deployment
+        applied = reconcile_once(metrics_url, kube)
+        assert (VA_NS, VA_NAME, desired) in applied
+        dep = kube.get(
+            f"/apis/apps/v1/namespaces/{VA_NS}/deployments/{VA_NAME}").json()
+        assert dep["spec"]["replicas"] == desired
+
+        # the controller's actuator reads REAL deployment replicas: the
+        # current_replicas gauge converges to what HPA applied
+        def current_follows():
+            text = httpx.get(metrics_url, timeout=5).text
+            for line in text.splitlines():
+                if line.startswith("inferno_current_replicas") and VA_NAME in line:
+                    if float(line.rsplit(" ", 1)[-1]) == float(desired):
+                        return True
+            return None
+
+        _wait_for(current_follows, 30, interval=2.0,
+                  desc="current_replicas gauge convergence")
