@@ -82,10 +82,11 @@ class LocalStack:
     """Owns the four processes + seeded objects."""
 
     def __init__(self, interval: str = "5s", backend: str = "auto",
-                 quiet: bool = False):
+                 quiet: bool = False, with_hpa: bool = False):
         self.interval = interval
         self.backend = backend
         self.quiet = quiet
+        self.with_hpa = with_hpa
         self.procs: list[subprocess.Popen] = []
         self.kube: httpx.Client | None = None
         self.api_port = self.emu_port = self.prom_port = self.metrics_port = 0
@@ -199,6 +200,19 @@ class LocalStack:
             raise RuntimeError("controller never became ready")
         self._log(f"controller ready; metrics/probes on :{self.metrics_port}")
 
+        # 6. optional HPA stand-in: closes the external-metric actuation loop
+        # (inferno_desired_replicas -> Deployment replicas)
+        if self.with_hpa:
+            p = subprocess.Popen(
+                [sys.executable, "-m", "inferno_amd.testing.hpa",
+                 "--metrics-url",
+                 f"http://127.0.0.1:{self.metrics_port}/metrics",
+                 "--kube-url", f"http://127.0.0.1:{self.api_port}",
+                 "--interval", "2"],
+                stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL, cwd=REPO)
+            self.procs.append(p)
+            self._log("HPA stand-in running (external-metric actuation loop)")
+
     def drive_load(self, seconds: float, concurrency: int = 6) -> int:
         emu = f"http://127.0.0.1:{self.emu_port}"
         stop = time.time() + seconds
@@ -255,9 +269,13 @@ def main() -> int:
                    help="drive load, wait for an optimized VA status, exit")
     p.add_argument("--interval", default="5s")
     p.add_argument("--backend", default="auto", choices=["auto", "gpu", "cpu"])
+    p.add_argument("--with-hpa", action="store_true",
+                   help="run the HPA stand-in so recommendations actually "
+                        "scale the Deployment")
     args = p.parse_args()
 
-    stack = LocalStack(interval=args.interval, backend=args.backend)
+    stack = LocalStack(interval=args.interval, backend=args.backend,
+                       with_hpa=args.with_hpa)
     try:
         stack.up()
         if args.smoke:
