@@ -116,10 +116,16 @@ def main() -> None:
 
     def watch_loop():
         """Create-only VA watch (the reference's event filter: Create -> true,
-        Update/Delete/Generic -> false, controller.go:456-487)."""
+        Update/Delete/Generic -> false, controller.go:456-487). Tracks the
+        last seen resourceVersion across the periodic reconnects so old
+        events are not replayed (which would fire spurious wake-ups)."""
+        rv = ""
         while not stop.is_set():
             try:
-                for etype, _va in kube.watch_events(timeout_seconds=55):
+                for etype, va in kube.watch_events(resource_version=rv,
+                                                   timeout_seconds=55):
+                    if va.resourceVersion:
+                        rv = va.resourceVersion
                     if etype == "ADDED":
                         wake.set()
                     if stop.is_set():

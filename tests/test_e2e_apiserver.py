@@ -414,3 +414,38 @@ class TestNodeInventoryOverHttp:
         inv = collector.collect_inventory_k8s(kube)
         assert inv["amd.com"]["MI355X"]["count"] == 4
         assert inv["nvidia.com"]["A100"]["count"] == 2
+
+
+class TestWatchResume:
+    def test_resume_skips_replayed_events(self, kubeapi):
+        """A watch opened at the store's current resourceVersion must not
+        replay history (the controller's reconnect path relies on this to
+        avoid spurious wake-ups)."""
+        client, port = kubeapi
+        # create one VA to have history, then note the current rv
+        TestApiMachinery()._mk_va(client, "va-resume-old")
+        lst = client.get("/apis/llmd.ai/v1alpha1/variantautoscalings").json()
+        rv = lst["metadata"]["resourceVersion"]
+
+        events = []
+
+        def consume():
+            with httpx.Client(base_url=f"http://127.0.0.1:{port}",
+                              timeout=15.0) as c:
+                with c.stream(
+                    "GET", "/apis/llmd.ai/v1alpha1/variantautoscalings",
+                    params={"watch": "1", "timeoutSeconds": "4",
+                            "resourceVersion": rv},
+                ) as r:
+                    for line in r.iter_lines():
+                        if line:
+                            events.append(json.loads(line))
+
+        t = threading.Thread(target=consume)
+        t.start()
+        time.sleep(0.5)
+        TestApiMachinery()._mk_va(client, "va-resume-new")
+        t.join(timeout=10)
+        names = [e["object"]["metadata"]["name"] for e in events]
+        assert "va-resume-new" in names
+        assert "va-resume-old" not in names  # history not replayed
