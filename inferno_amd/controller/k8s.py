@@ -294,13 +294,17 @@ class HttpKube:
                 yield etype, api.va_from_json(obj)
 
     def watch_configmap_events(self, namespace: str, names: set[str],
-                               timeout_seconds: int = 60):
+                               timeout_seconds: int = 60,
+                               resource_version: str = ""):
         """Stream watch events for the controller's ConfigMaps (the
         reference's Watches(ConfigMap) registration, controller.go:456-487).
-        Yields (event_type, name)."""
+        Yields (event_type, name, resourceVersion) — callers thread the last
+        resourceVersion back in on reconnect so history is not replayed."""
         import json as _json
 
         params = {"watch": "1", "timeoutSeconds": str(timeout_seconds)}
+        if resource_version:
+            params["resourceVersion"] = resource_version
         with self._client.stream(
             "GET", f"/api/v1/namespaces/{namespace}/configmaps", params=params,
             timeout=timeout_seconds + 10,
@@ -313,9 +317,11 @@ class HttpKube:
                     evt = _json.loads(line)
                 except ValueError:
                     continue
-                name = ((evt.get("object") or {}).get("metadata") or {}).get("name", "")
+                meta = (evt.get("object") or {}).get("metadata") or {}
+                name = meta.get("name", "")
+                rv = str(meta.get("resourceVersion", "") or "")
                 if name in names:
-                    yield evt.get("type", ""), name
+                    yield evt.get("type", ""), name, rv
 
     def set_owner_reference(self, va: api.VariantAutoscaling, deploy: Deployment) -> None:
         path = f"/apis/{api.GROUP}/{api.VERSION}/namespaces/{va.namespace}/{api.PLURAL}/{va.name}"

@@ -139,11 +139,15 @@ def main() -> None:
         from .reconciler import ACCELERATOR_CM, SERVICE_CLASS_CM, WVA_CONFIG_CM
 
         names = {ACCELERATOR_CM, SERVICE_CLASS_CM, WVA_CONFIG_CM}
+        rv = ""
         while not stop.is_set():
             try:
-                for _etype, _name in kube.watch_configmap_events(
-                    args.configmap_namespace, names, timeout_seconds=55
+                for _etype, _name, ev_rv in kube.watch_configmap_events(
+                    args.configmap_namespace, names, timeout_seconds=55,
+                    resource_version=rv,
                 ):
+                    if ev_rv:
+                        rv = ev_rv
                     wake.set()
                     if stop.is_set():
                         return

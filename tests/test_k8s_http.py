@@ -227,11 +227,13 @@ class TestConfigMapWatch:
     def test_filters_to_controller_configmaps(self):
         lines = [
             json.dumps({"type": "MODIFIED", "object": {
-                "metadata": {"name": "service-classes-config"}}}),
+                "metadata": {"name": "service-classes-config",
+                             "resourceVersion": "5"}}}),
             json.dumps({"type": "MODIFIED", "object": {
-                "metadata": {"name": "unrelated-cm"}}}),
+                "metadata": {"name": "unrelated-cm", "resourceVersion": "6"}}}),
             json.dumps({"type": "ADDED", "object": {
-                "metadata": {"name": "accelerator-unit-costs"}}}),
+                "metadata": {"name": "accelerator-unit-costs",
+                             "resourceVersion": "7"}}}),
         ]
 
         def handler(req):
@@ -241,8 +243,16 @@ class TestConfigMapWatch:
         events = list(make_kube(handler).watch_configmap_events(
             "ns", {"service-classes-config", "accelerator-unit-costs"},
             timeout_seconds=1))
-        assert events == [("MODIFIED", "service-classes-config"),
-                          ("ADDED", "accelerator-unit-costs")]
+        assert events == [("MODIFIED", "service-classes-config", "5"),
+                          ("ADDED", "accelerator-unit-costs", "7")]
+
+    def test_resource_version_param_passed(self):
+        def handler(req):
+            assert req.url.params.get("resourceVersion") == "41"
+            return httpx.Response(200, text="")
+
+        list(make_kube(handler).watch_configmap_events(
+            "ns", {"x"}, timeout_seconds=1, resource_version="41"))
 
 
 class TestLeaderFailClosed:
