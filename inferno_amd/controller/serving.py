@@ -133,13 +133,19 @@ class _TLSServer(ThreadingHTTPServer):
     daemon_threads = True
     watcher: Optional[_CertWatcher] = None
 
-    def get_request(self):
-        sock, addr = self.socket.accept()
+    def finish_request(self, request, client_address):
+        # wrap per-connection with the *current* context so hot-reloaded
+        # certs take effect without rebinding the listener. Done HERE (in
+        # the per-connection worker thread, not get_request) so a slow or
+        # malicious client's handshake cannot stall the accept loop; a
+        # failed handshake raises into handle_error and only kills this
+        # connection.
         if self.watcher is not None:
-            # wrap per-connection with the *current* context so hot-reloaded
-            # certs take effect without rebinding the listener
-            sock = self.watcher.context.wrap_socket(sock, server_side=True)
-        return sock, addr
+            request = self.watcher.context.wrap_socket(request, server_side=True)
+        super().finish_request(request, client_address)
+
+    def handle_error(self, request, client_address):
+        pass  # per-connection TLS/parse failures are not server errors
 
 
 class ProbeServer:
