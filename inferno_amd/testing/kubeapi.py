@@ -167,10 +167,17 @@ class Store:
             has_status="status" in (version.get("subresources") or {}),
         )
 
+    # cap on retained watch history; resuming from an rv older than the
+    # window behaves like the real apiserver's compaction (events are simply
+    # gone — clients re-list). Bounds memory for long-lived `stack` runs.
+    MAX_EVENTS = 50_000
+
     # -- CRUD ----------------------------------------------------------
     def _emit(self, etype: str, plural: str, obj: dict) -> None:
         self.events.append((int(obj["metadata"]["resourceVersion"]), etype, plural,
                             copy.deepcopy(obj)))
+        if len(self.events) > self.MAX_EVENTS:
+            del self.events[: len(self.events) - self.MAX_EVENTS]
 
     def create(self, plural: str, ns: str, body: dict) -> dict:
         kind = self.kinds[plural]
