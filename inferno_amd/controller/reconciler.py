@@ -36,6 +36,23 @@ def _now_iso() -> str:
     return datetime.now(timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
 
 
+def _gpu_ready() -> bool:
+    """A GPU re-probe may only swap the engine back when a device is visible
+    AND the native library loads — otherwise the swapped-in engine fails on
+    the next tick and churns through the fallback path repeatedly."""
+    try:
+        from ..engine.engine import _gpu_available
+
+        if not _gpu_available():
+            return False
+        from ..ops.sweep import load_library
+
+        load_library(allow_build=False)
+        return True
+    except Exception:
+        return False
+
+
 def parse_go_duration(s: str) -> float:
     """Parse a Go-style duration string ("60s", "1m30s", "500ms") to seconds."""
     import re
@@ -201,6 +218,8 @@ class Reconciler:
             return
         self._degraded_ticks = 0
         try:
+            if not _gpu_ready():
+                return
             probe = SweepEngine(backend="gpu")
             if probe.backend == "gpu":
                 self.engine = probe

@@ -592,6 +592,7 @@ class TestGpuReprobe:
             def __init__(self, backend="auto", device="cuda"):
                 probes["n"] += 1
 
+        monkeypatch.setattr(rmod, "_gpu_ready", lambda: True)
         monkeypatch.setattr(rmod, "SweepEngine", FakeEngine)
         for _ in range(7):
             rec.reconcile()
@@ -606,6 +607,7 @@ class TestGpuReprobe:
 
         import inferno_amd.controller.reconciler as rmod
 
+        monkeypatch.setattr(rmod, "_gpu_ready", lambda: True)
         real_engine = type(rec.engine)
 
         class GpuOkEngine:
