@@ -687,3 +687,13 @@ class TestInventoryCollection:
         monkeypatch.setenv("WVA_LIMITED_MODE", "true")
         result = rec.reconcile()
         assert result.processed == 1  # no inventory -> unlimited as before
+
+    def test_no_gpu_no_swap(self):
+        """On a box with no visible device, the re-probe never swaps a
+        broken GPU engine in (would churn the fallback path)."""
+        kube, prom, em, reg, rec = build_world(arrival_per_sec=2.0)
+        rec.engine.degraded_from_gpu = True
+        rec.gpu_reprobe_interval = 1
+        rec._maybe_reprobe_gpu()  # container has no GPU -> _gpu_ready False
+        assert rec.engine.backend == "cpu"
+        assert rec.engine.degraded_from_gpu is True
