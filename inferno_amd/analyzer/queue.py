@@ -229,7 +229,11 @@ def build_service_rates(cfg: Configuration, req: RequestSize) -> np.ndarray:
     if req.avg_input_tokens == 0 and req.avg_output_tokens == 1:
         num_decode = 1
     decode = (np.float32(parms.decode.alpha) + np.float32(parms.decode.beta) * n).astype(np.float32)
-    serv = (n / (prefill + np.float32(num_decode) * decode)).astype(np.float32)
+    # degenerate perf parms can yield a zero/negative service time; the
+    # resulting inf/nan rate is rejected by the callers' validity checks
+    # (AnalyzerError / infeasible cell) — silence only the expected warning
+    with np.errstate(divide="ignore", invalid="ignore"):
+        serv = (n / (prefill + np.float32(num_decode) * decode)).astype(np.float32)
     return serv
 
 
