@@ -35,8 +35,6 @@ def build_app():
     )
     avg_out = float(os.environ.get("AVG_OUTPUT_TOKENS", "64"))
     registry = make_registry(sim, model_name, namespace)
-    app = FastAPI(title="inferno-amd vllm emulator")
-    app.state.sim = sim
 
     async def run_loop():
         # advance the simulated scheduler in real time
@@ -48,13 +46,18 @@ def build_app():
             else:
                 await asyncio.sleep(sim.decode_time_s)
 
-    @app.on_event("startup")
-    async def _start():
-        app.state.loop_task = asyncio.create_task(run_loop())
+    from contextlib import asynccontextmanager
 
-    @app.on_event("shutdown")
-    async def _stop():
-        app.state.loop_task.cancel()
+    @asynccontextmanager
+    async def lifespan(app_):
+        app_.state.loop_task = asyncio.create_task(run_loop())
+        try:
+            yield
+        finally:
+            app_.state.loop_task.cancel()
+
+    app = FastAPI(title="inferno-amd vllm emulator", lifespan=lifespan)
+    app.state.sim = sim
 
     @app.post("/v1/chat/completions")
     async def chat_completions(request: Request):
