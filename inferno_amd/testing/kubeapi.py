@@ -479,6 +479,30 @@ def build_app(store: Store, valid_tokens: Optional[set[str]] = None):
                            plural: str, name: str):
         return await handle(request, plural, ns, name, subresource="status")
 
+    @app.api_route("/apis/apps/v1/namespaces/{ns}/deployments/{name}/scale",
+                   methods=["GET", "PUT", "PATCH"])
+    async def deployment_scale(request: Request, ns: str, name: str):
+        """The autoscaling/v1 Scale subresource real HPA controllers use."""
+        obj = store.get("deployments", ns, name)
+        if obj is None:
+            return status_err(404, "NotFound", f"deployments/{name} not found")
+        if request.method in ("PUT", "PATCH"):
+            body = json.loads(await request.body() or b"{}")
+            replicas = int(((body.get("spec") or {}).get("replicas", 0)) or 0)
+            try:
+                obj = store.patch("deployments", ns, name,
+                                  {"spec": {"replicas": replicas}})
+            except ConflictError as e:
+                return status_err(409, "Conflict", str(e))
+        return JSONResponse({
+            "apiVersion": "autoscaling/v1",
+            "kind": "Scale",
+            "metadata": {"name": name, "namespace": ns,
+                         "resourceVersion": obj["metadata"]["resourceVersion"]},
+            "spec": {"replicas": (obj.get("spec") or {}).get("replicas", 0)},
+            "status": {"replicas": (obj.get("status") or {}).get("replicas", 0)},
+        })
+
     @app.get("/healthz")
     async def healthz():
         return JSONResponse({"ok": True})

@@ -449,3 +449,24 @@ class TestWatchResume:
         names = [e["object"]["metadata"]["name"] for e in events]
         assert "va-resume-new" in names
         assert "va-resume-old" not in names  # history not replayed
+
+
+class TestScaleSubresource:
+    def test_scale_get_and_patch(self, kubeapi):
+        """autoscaling/v1 Scale subresource — what a REAL HPA controller
+        talks to when scaling the Deployment."""
+        client, _ = kubeapi
+        r = _apply(client, {
+            "apiVersion": "apps/v1", "kind": "Deployment",
+            "metadata": {"name": "scaled-dep", "namespace": "machinery"},
+            "spec": {"replicas": 1}})
+        assert r.status_code == 201
+        path = "/apis/apps/v1/namespaces/machinery/deployments/scaled-dep/scale"
+        got = client.get(path).json()
+        assert got["kind"] == "Scale" and got["spec"]["replicas"] == 1
+        r = client.patch(path, json={"spec": {"replicas": 5}},
+                         headers={"Content-Type": "application/merge-patch+json"})
+        assert r.status_code == 200 and r.json()["spec"]["replicas"] == 5
+        dep = client.get(
+            "/apis/apps/v1/namespaces/machinery/deployments/scaled-dep").json()
+        assert dep["spec"]["replicas"] == 5
