@@ -334,6 +334,17 @@ __device__ ChainOut chain_eval(double lam, const ChainGeom &g, double logsN, int
       // SQ_WAIT_ANY was 50% of wave cycles). Arithmetic order is unchanged
       // (bit-identical results).
       int j = j0 + 1;
+      for (; j + 7 <= jend; j += 8) {
+        double f[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) f[q] = (double)g.inv_s_t[(j + q) * NT + c];
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          w *= lam * f[q];
+          head_sum += w;
+          head_n_sum += (double)(n0c + j + q) * w;
+        }
+      }
       for (; j + 3 <= jend; j += 4) {
         const double f0 = (double)g.inv_s_t[j * NT + c];
         const double f1 = (double)g.inv_s_t[(j + 1) * NT + c];
