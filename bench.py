@@ -53,6 +53,14 @@ def _parse_args():
         "per accelerator type (0 = unlimited argmin, the default)",
     )
     p.add_argument(
+        "--scaling",
+        choices=["weak", "strong"],
+        default="weak",
+        help="weak (default): each GPU owns a models-per-gpu shard, fleet "
+        "grows with N. strong: the TOTAL fleet is fixed at models-per-gpu "
+        "and split across ranks",
+    )
+    p.add_argument(
         "--preset",
         choices=["config2", "config3", "config4", "config5"],
         default=None,
@@ -127,8 +135,9 @@ def main():
             local_rank = int(os.environ.get("LOCAL_RANK", rank))
             torch.cuda.set_device(local_rank % n_dev)
 
+    fleet_world = 1 if args.scaling == "strong" else world
     spec, args.models_per_gpu = preset_fleet(
-        args.preset, args.models_per_gpu, world, args.seed
+        args.preset, args.models_per_gpu, fleet_world, args.seed
     )
     if args.analyzer:
         spec.optimizer.analyzer = args.analyzer
@@ -271,7 +280,7 @@ def main():
         "warmup": args.warmup,
         "ms_per_step": round(ms_per_step, 3),
         "higher_is_better": True,
-        "scaling": "weak",
+        "scaling": args.scaling,
         "vs_baseline": None,
         "dtype": "fp64",
         "data": "synthetic (random-init perf profiles, bursty Poisson trace)",

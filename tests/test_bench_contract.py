@@ -117,3 +117,25 @@ class TestLocalRankMapping:
                               (8, list(range(8)))):
             got = [lr % n_dev for lr in range(8)]
             assert got == expect
+
+
+class TestStrongScaling:
+    def test_strong_mode_fixed_total_fleet(self):
+        """--scaling strong: the TOTAL fleet stays models-per-gpu as world
+        grows (the driver computes the strong-scaling curve from this)."""
+        env = dict(os.environ)
+        env["INFERNO_DIST_BACKEND"] = "gloo"
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29928",
+             os.path.join(REPO, "bench.py"), "--gpus", "2", "--backend", "cpu",
+             "--scaling", "strong", "--models-per-gpu", "8",
+             "--steps", "2", "--warmup", "1"],
+            capture_output=True, text=True, timeout=600, cwd=REPO, env=env,
+        )
+        assert proc.returncode == 0, proc.stderr[-2000:]
+        d = json.loads([l for l in proc.stdout.splitlines() if l.startswith("{")][0])
+        assert d["scaling"] == "strong"
+        # total cells = 8 models x 3 accs regardless of world size
+        assert d["config"]["cells_per_step"] == 8 * 3
