@@ -45,6 +45,10 @@ def gpu_health_probe() -> dict:
 def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--metrics-port", type=int, default=8443)
+    p.add_argument("--probe-port", type=int, default=0,
+                   help="serve healthz/readyz on a separate plain-HTTP port "
+                        "(controller-runtime's split layout: probes on 8081, "
+                        "secured metrics on 8443); 0 = same port as metrics")
     p.add_argument("--metrics-cert-dir", default=os.environ.get("METRICS_CERT_DIR", ""),
                    help="dir with tls.crt/tls.key; enables HTTPS + hot reload "
                         "(ref cmd/main.go:122-155 certwatcher)")
@@ -80,10 +84,13 @@ def main() -> None:
         client_ca=args.metrics_client_ca or None,
         auth=auth,
     )
+    if args.probe_port:
+        # plain-HTTP probe-only listener for the kubelet (no /metrics)
+        ProbeServer(args.probe_port, state, expose_metrics=False)
     logger.info(
         "metrics/probe server started",
         extra={"kv": {"port": probe_server.port, "tls": probe_server.tls,
-                      "auth": auth.enabled}},
+                      "auth": auth.enabled, "probe_port": args.probe_port or None}},
     )
     cm = kube.get_configmap(args.configmap_namespace, WVA_CONFIG_CM) or {}
     prom_cfg = collector.prometheus_config_from_env(cm)

@@ -159,10 +159,12 @@ class ProbeServer:
         client_ca: Optional[str] = None,
         auth: Optional[MetricsAuth] = None,
         bind: str = "0.0.0.0",
+        expose_metrics: bool = True,
     ):
         from prometheus_client import generate_latest
 
         auth = auth or MetricsAuth()
+        self.expose_metrics = expose_metrics
         watcher: Optional[_CertWatcher] = None
         if cert_dir:
             cert = os.path.join(cert_dir, "tls.crt")
@@ -181,6 +183,12 @@ class ProbeServer:
                         self.send_response(503)
                         self.end_headers()
                 elif self.path == "/metrics":
+                    if not expose_metrics:
+                        # probe-only listener (split-port layout, ref
+                        # cmd/main.go: probes on 8081, metrics on 8443)
+                        self.send_response(404)
+                        self.end_headers()
+                        return
                     if not auth.check(self.headers.get("Authorization", "")):
                         self.send_response(401)
                         self.send_header("WWW-Authenticate", "Bearer")

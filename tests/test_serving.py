@@ -143,3 +143,18 @@ class TestAuth:
         assert auth.check("Bearer good") is True
         assert auth.check("Bearer bad") is False
         assert auth.check("") is False
+
+
+class TestProbeOnlyListener:
+    def test_split_port_layout(self):
+        """Probe-only listener (controller-runtime's 8081/8443 split): serves
+        healthz/readyz but NOT /metrics."""
+        srv = ProbeServer(0, {"ready": True}, bind="127.0.0.1",
+                          expose_metrics=False)
+        try:
+            base = f"http://127.0.0.1:{srv.port}"
+            assert httpx.get(f"{base}/healthz").status_code == 200
+            assert httpx.get(f"{base}/readyz").status_code == 200
+            assert httpx.get(f"{base}/metrics").status_code == 404
+        finally:
+            srv.shutdown()
