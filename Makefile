@@ -13,6 +13,10 @@ build:
 test:
 	$(PYTHON) -m pytest tests/ -x -q -m "not gpu"
 
+## CPU suite across 4 workers (developer loop; ~40% faster)
+test-fast:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu" -n 4
+
 ## GPU differential + smoke tests (requires MI355X)
 test-gpu:
 	$(PYTHON) -m pytest tests/ -x -q -m gpu
