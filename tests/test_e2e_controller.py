@@ -8,10 +8,8 @@ cross-checks, :519 scale-in at idle, conditions/ownerRef/lease assertions)
 """
 from __future__ import annotations
 
-import json
 import os
 import signal
-import socket
 import subprocess
 import sys
 import threading

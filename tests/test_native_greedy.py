@@ -5,7 +5,7 @@ entry point is pure CPU)."""
 import numpy as np
 import pytest
 
-from inferno_amd.config import OptimizerSpec, SaturationPolicy, ServerLoadSpec
+from inferno_amd.config import SaturationPolicy, ServerLoadSpec
 from inferno_amd.core.system import System
 from inferno_amd.engine import SweepEngine
 from inferno_amd.engine.fastpath import FastSweep
